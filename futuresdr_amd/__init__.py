@@ -1,0 +1,349 @@
+"""futuresdr_amd — MI355X-native FutureSDR streaming-DSP hot path.
+
+Python plumbing over the product C-ABI (include/futuresdr_hip.h,
+implemented in csrc/futuresdr_hip.hip). The compute path is the HIP
+extension; there is NO CPU fallback here — if the shared library or a HIP
+device is missing, calls raise. The CPU restatement used for parity lives
+in oracle/ (test infrastructure) and is never imported by this package.
+"""
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libfutursdr_hip.so")
+
+CF32 = np.dtype(np.complex64)
+
+INSUFFICIENT_INPUT = 0
+INSUFFICIENT_OUTPUT = 1
+BOTH_SUFFICIENT = 2
+
+OK = 0
+ERR_NO_GPU = 1
+ERR_HIP = 2
+ERR_INVALID = 3
+ERR_UNSUPPORTED = 4
+
+
+class FsdrError(RuntimeError):
+    pass
+
+
+class _Result(ctypes.Structure):
+    _fields_ = [
+        ("consumed", ctypes.c_size_t),
+        ("produced", ctypes.c_size_t),
+        ("status", ctypes.c_int),
+    ]
+
+
+def build():
+    """Compile the gfx950 HIP extension in-tree (hipcc, ~5 s)."""
+    subprocess.run(["make", "-s", "-C", _DIR], check=True)
+
+
+_lib = None
+
+
+def _load():
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_SO):
+        raise ImportError(
+            f"futuresdr_amd HIP extension missing: {_SO} not built. "
+            "Run futuresdr_amd.build() (or make -C futuresdr_amd). "
+            "There is no CPU fallback — this is the GPU product path.")
+    lib = ctypes.CDLL(_SO)
+    sz = ctypes.c_size_t
+    vp = ctypes.c_void_p
+    f32p = ctypes.POINTER(ctypes.c_float)
+    rp = ctypes.POINTER(_Result)
+
+    lib.fsdr_last_error.restype = ctypes.c_char_p
+    lib.fsdr_version.restype = ctypes.c_char_p
+    lib.fsdr_device_count.restype = ctypes.c_int
+    lib.fsdr_set_device.argtypes = [ctypes.c_int]
+    lib.fsdr_fir_cf32_create.restype = vp
+    lib.fsdr_fir_cf32_create.argtypes = [f32p, sz]
+    lib.fsdr_fir_f32_create.restype = vp
+    lib.fsdr_fir_f32_create.argtypes = [f32p, sz]
+    lib.fsdr_decim_fir_cf32_create.restype = vp
+    lib.fsdr_decim_fir_cf32_create.argtypes = [sz, f32p, sz]
+    lib.fsdr_resamp_cf32_create.restype = vp
+    lib.fsdr_resamp_cf32_create.argtypes = [sz, sz, f32p, sz]
+    lib.fsdr_fft_cf32_create.restype = vp
+    lib.fsdr_fft_cf32_create.argtypes = [sz, ctypes.c_int, ctypes.c_int, f32p]
+    lib.fsdr_mag2_create.restype = vp
+    lib.fsdr_filter_length.restype = sz
+    lib.fsdr_filter_length.argtypes = [vp]
+    lib.fsdr_filter_host.restype = ctypes.c_int
+    lib.fsdr_filter_host.argtypes = [vp, vp, sz, vp, sz, rp]
+    lib.fsdr_filter_dev.restype = ctypes.c_int
+    lib.fsdr_filter_dev.argtypes = [vp, vp, sz, vp, sz, vp, rp]
+    lib.fsdr_filter_destroy.argtypes = [vp]
+    lib.fsdr_cmul_dev.restype = ctypes.c_int
+    lib.fsdr_cmul_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
+                                  ctypes.POINTER(sz)]
+    lib.fsdr_cmul_host.restype = ctypes.c_int
+    lib.fsdr_cmul_host.argtypes = [vp, sz, vp, sz, vp, sz,
+                                   ctypes.POINTER(sz)]
+    lib.fsdr_dev_alloc.restype = ctypes.c_int
+    lib.fsdr_dev_alloc.argtypes = [ctypes.POINTER(vp), sz]
+    lib.fsdr_dev_free.argtypes = [vp]
+    lib.fsdr_memcpy_h2d.restype = ctypes.c_int
+    lib.fsdr_memcpy_h2d.argtypes = [vp, vp, sz]
+    lib.fsdr_memcpy_d2h.restype = ctypes.c_int
+    lib.fsdr_memcpy_d2h.argtypes = [vp, vp, sz]
+    lib.fsdr_fill_uniform_cf32.restype = ctypes.c_int
+    lib.fsdr_fill_uniform_cf32.argtypes = [vp, sz, ctypes.c_uint64,
+                                           ctypes.c_uint64, vp]
+    lib.fsdr_kaiser_beta.restype = ctypes.c_double
+    lib.fsdr_kaiser_beta.argtypes = [ctypes.c_double]
+    lib.fsdr_kaiser_window.restype = None
+    lib.fsdr_kaiser_window.argtypes = [sz, ctypes.c_double,
+                                       ctypes.POINTER(ctypes.c_double)]
+    lib.fsdr_firdes_kaiser_lowpass_f32.restype = sz
+    lib.fsdr_firdes_kaiser_lowpass_f32.argtypes = [
+        ctypes.c_double, ctypes.c_double, ctypes.c_double, f32p, sz]
+    lib.fsdr_firdes_lowpass_kaiser_n_f32.restype = ctypes.c_int
+    lib.fsdr_firdes_lowpass_kaiser_n_f32.argtypes = [
+        sz, ctypes.c_double, ctypes.c_double, f32p]
+    lib.fsdr_chain_create.restype = vp
+    lib.fsdr_chain_create.argtypes = [f32p, sz, f32p, sz, sz, sz]
+    lib.fsdr_chain_run_dev.restype = ctypes.c_int
+    lib.fsdr_chain_run_dev.argtypes = [vp, vp, sz, vp, sz, vp, sz, vp,
+                                       ctypes.POINTER(sz),
+                                       ctypes.POINTER(sz)]
+    lib.fsdr_chain_destroy.argtypes = [vp]
+    lib.fsdr_synchronize.restype = ctypes.c_int
+    lib.fsdr_ring_create.restype = vp
+    lib.fsdr_ring_create.argtypes = [sz, sz, sz, sz]
+    lib.fsdr_ring_writer_acquire.restype = ctypes.c_int
+    lib.fsdr_ring_writer_acquire.argtypes = [vp, ctypes.POINTER(vp),
+                                             ctypes.POINTER(sz)]
+    lib.fsdr_ring_writer_commit.restype = ctypes.c_int
+    lib.fsdr_ring_writer_commit.argtypes = [vp, sz]
+    lib.fsdr_ring_reader_acquire.restype = ctypes.c_int
+    lib.fsdr_ring_reader_acquire.argtypes = [vp, ctypes.POINTER(vp),
+                                             ctypes.POINTER(sz)]
+    lib.fsdr_ring_reader_release.restype = ctypes.c_int
+    lib.fsdr_ring_reader_release.argtypes = [vp]
+    lib.fsdr_ring_destroy.argtypes = [vp]
+    _lib = lib
+    return lib
+
+
+def lib():
+    return _load()
+
+
+def exported_symbols():
+    """Every symbol include/futuresdr_hip.h declares (for the CPU test)."""
+    hdr = os.path.join(_DIR, "..", "include", "futuresdr_hip.h")
+    import re
+    syms = []
+    with open(hdr) as f:
+        for m in re.finditer(r"\b(fsdr_\w+)\s*\(", f.read()):
+            syms.append(m.group(1))
+    return sorted(set(syms))
+
+
+def _check(rc):
+    if rc != OK:
+        raise FsdrError(f"fsdr error {rc}: "
+                        f"{_load().fsdr_last_error().decode()}")
+
+
+def _f32(a):
+    a = np.ascontiguousarray(a, np.float32)
+    return a, a.ctypes.data_as(ctypes.POINTER(ctypes.c_float))
+
+
+def version():
+    return _load().fsdr_version().decode()
+
+
+def device_count():
+    return _load().fsdr_device_count()
+
+
+def set_device(i):
+    _check(_load().fsdr_set_device(i))
+
+
+def synchronize():
+    _check(_load().fsdr_synchronize())
+
+
+class Filter:
+    """GPU filter handle mirroring futuredsp::Filter (lib.rs:48-68)."""
+
+    ITEM_IN = CF32
+    ITEM_OUT = CF32
+
+    def __init__(self, handle):
+        if not handle:
+            raise FsdrError("filter create failed: "
+                            f"{_load().fsdr_last_error().decode()}")
+        self._h = handle
+
+    @property
+    def length(self):
+        return _load().fsdr_filter_length(self._h)
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _load().fsdr_filter_destroy(self._h)
+            self._h = None
+
+    def filter(self, inp, n_out):
+        """Host-span path: numpy in -> (out, consumed, produced, status)."""
+        lib = _load()
+        inp = np.ascontiguousarray(inp, self.ITEM_IN)
+        out = np.zeros(n_out, self.ITEM_OUT)
+        r = _Result()
+        _check(lib.fsdr_filter_host(
+            self._h, ctypes.c_void_p(inp.ctypes.data), inp.size,
+            ctypes.c_void_p(out.ctypes.data), out.size, ctypes.byref(r)))
+        return out[: r.produced], r.consumed, r.produced, r.status
+
+    def filter_dev(self, d_in, n_in, d_out, n_out, stream=None):
+        """Device-pointer path (ints = device addresses). Async."""
+        lib = _load()
+        r = _Result()
+        _check(lib.fsdr_filter_dev(self._h, ctypes.c_void_p(d_in), n_in,
+                                   ctypes.c_void_p(d_out), n_out,
+                                   ctypes.c_void_p(stream or 0),
+                                   ctypes.byref(r)))
+        return r.consumed, r.produced, r.status
+
+
+class Fir(Filter):
+    """Fir block core — src/blocks/fir.rs + futuredsp fir.rs (cf32 x f32)."""
+
+    def __init__(self, taps):
+        self._taps_keep, p = _f32(taps)
+        super().__init__(_load().fsdr_fir_cf32_create(p,
+                                                      self._taps_keep.size))
+
+
+class FirF32(Filter):
+    ITEM_IN = np.dtype(np.float32)
+    ITEM_OUT = np.dtype(np.float32)
+
+    def __init__(self, taps):
+        self._taps_keep, p = _f32(taps)
+        super().__init__(_load().fsdr_fir_f32_create(p, self._taps_keep.size))
+
+
+class DecimFir(Filter):
+    def __init__(self, decimation, taps):
+        self._taps_keep, p = _f32(taps)
+        super().__init__(_load().fsdr_decim_fir_cf32_create(
+            decimation, p, self._taps_keep.size))
+
+
+class Resampler(Filter):
+    def __init__(self, interp, decim, taps):
+        self._taps_keep, p = _f32(taps)
+        super().__init__(_load().fsdr_resamp_cf32_create(
+            interp, decim, p, self._taps_keep.size))
+
+
+class Fft(Filter):
+    def __init__(self, length, inverse=False, fft_shift=False,
+                 normalize=None):
+        np_ = None
+        if normalize is not None:
+            np_ = ctypes.pointer(ctypes.c_float(normalize))
+        super().__init__(_load().fsdr_fft_cf32_create(
+            length, int(inverse), int(fft_shift), np_))
+
+
+class Mag2(Filter):
+    ITEM_OUT = np.dtype(np.float32)
+
+    def __init__(self):
+        super().__init__(_load().fsdr_mag2_create())
+
+
+class Chain:
+    """Fused Fir -> DecimFir -> Fft pipeline (the bench hot path)."""
+
+    def __init__(self, taps1, taps2, decim, fft_len):
+        lib = _load()
+        self._t1, p1 = _f32(taps1)
+        self._t2, p2 = _f32(taps2)
+        self.decim, self.fft_len = decim, fft_len
+        self._h = lib.fsdr_chain_create(p1, self._t1.size, p2, self._t2.size,
+                                        decim, fft_len)
+        if not self._h:
+            raise FsdrError("chain create failed: "
+                            f"{lib.fsdr_last_error().decode()}")
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _load().fsdr_chain_destroy(self._h)
+            self._h = None
+
+    def run_dev(self, d_in, n_in, d_out=0, out_cap=0, d_mag=0, mag_cap=0,
+                stream=None):
+        lib = _load()
+        cons = ctypes.c_size_t()
+        prod = ctypes.c_size_t()
+        _check(lib.fsdr_chain_run_dev(
+            self._h, ctypes.c_void_p(d_in), n_in, ctypes.c_void_p(d_out),
+            out_cap, ctypes.c_void_p(d_mag), mag_cap,
+            ctypes.c_void_p(stream or 0), ctypes.byref(cons),
+            ctypes.byref(prod)))
+        return cons.value, prod.value
+
+
+def kaiser_lowpass(cutoff, transition_bw, max_ripple):
+    """firdes::kaiser::lowpass<f32> — product host-side designer."""
+    lib = _load()
+    n = lib.fsdr_firdes_kaiser_lowpass_f32(cutoff, transition_bw,
+                                           max_ripple, None, 0)
+    out = np.zeros(n, np.float32)
+    lib.fsdr_firdes_kaiser_lowpass_f32(
+        cutoff, transition_bw, max_ripple,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), out.size)
+    return out
+
+
+def lowpass_kaiser_n(n_taps, beta, cutoff):
+    """firdes::lowpass over a kaiser(n_taps, beta) window (basic.rs:25-42)."""
+    lib = _load()
+    out = np.zeros(n_taps, np.float32)
+    _check(lib.fsdr_firdes_lowpass_kaiser_n_f32(
+        n_taps, beta, cutoff,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_float))))
+    return out
+
+
+def kaiser_beta(max_ripple):
+    return _load().fsdr_kaiser_beta(max_ripple)
+
+
+def cmul_host(a, b):
+    lib = _load()
+    a = np.ascontiguousarray(a, CF32)
+    b = np.ascontiguousarray(b, CF32)
+    m = min(a.size, b.size)
+    out = np.zeros(m, CF32)
+    mm = ctypes.c_size_t()
+    _check(lib.fsdr_cmul_host(ctypes.c_void_p(a.ctypes.data), a.size,
+                              ctypes.c_void_p(b.ctypes.data), b.size,
+                              ctypes.c_void_p(out.ctypes.data), out.size,
+                              ctypes.byref(mm)))
+    return out[: mm.value]
+
+
+def fill_uniform_dev(d_ptr, n, seed=0, offset=0, stream=None):
+    _check(_load().fsdr_fill_uniform_cf32(ctypes.c_void_p(d_ptr), n, seed,
+                                          offset,
+                                          ctypes.c_void_p(stream or 0)))

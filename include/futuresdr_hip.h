@@ -1,0 +1,191 @@
+/* futuresdr_hip.h — C-ABI of the MI355X-native FutureSDR streaming-DSP hot
+ * path. This is the drop-in boundary: plain pointers and
+ * sizes, no framework types. A FutureSDR maintainer binds these symbols from
+ * Rust with a plain `extern "C"` block.
+ *
+ * Contract mirrored, per entry point:
+ *  - fsdr_filter_host/_dev mirror `futuredsp::Filter::filter(&self, &[I],
+ *    &mut [O]) -> (usize, usize, ComputationStatus)` —
+ *    /root/reference/crates/futuredsp/src/lib.rs:48-68. Same
+ *    consumed/produced/status math as the cores they replace:
+ *      fir:    crates/futuredsp/src/fir.rs:52-91
+ *      decim:  crates/futuredsp/src/decimating_fir.rs:53-95
+ *      resamp: crates/futuredsp/src/polyphase_resampling_fir.rs:70-124
+ *      fft:    src/blocks/fft.rs:160-221 (m = min(in,out) rounded to len,
+ *              capped at 32*len; consumed == produced == m)
+ *      mag2:   src/blocks/apply.rs:100-131 (m = min(in,out))
+ *      cmul:   src/blocks/combine.rs:92-135 (m = min(in0,in1,out))
+ *  - fsdr_ring_* mirror the Slab stream-buffer circulation
+ *    (src/runtime/buffer/slab.rs:110-152,369-399) and the accelerator
+ *    buffer pattern (src/runtime/buffer/vulkan/{h2d,d2h}.rs,
+ *    src/runtime/buffer/wgpu): N pinned-host buffers cycling empty<->full
+ *    with a reserved-items history prefix.
+ *  - fsdr_chain_* is the fused pipeline runner (the `fsdr_chain_run` of
+ *    SURVEY.md §8b).
+ *
+ * Failure convention: int status + thread-local error string
+ * (fsdr_last_error). The Rust Filter trait has no error path (lib.rs:60-64);
+ * the ABI's failures are device/allocation errors only. All compute entry
+ * points fail with FSDR_ERR_NO_GPU when no HIP device is present — there is
+ * NO CPU fallback in the product (the CPU restatement lives in oracle/,
+ * which is test infrastructure).
+ *
+ * Threading: a handle may be used from one thread at a time (same as the
+ * reference, where one block owns its kernel). Distinct handles are
+ * independent. `stream` parameters take a hipStream_t (or NULL for the
+ * default stream) so callers (e.g. torch) can pass their own streams.
+ */
+#ifndef FUTURESDR_HIP_H
+#define FUTURESDR_HIP_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Complex32 — layout-identical to num_complex::Complex<f32> (re, im). */
+typedef struct { float re, im; } fsdr_cf32;
+
+/* ComputationStatus — crates/futuredsp/src/lib.rs:31-45 */
+typedef enum {
+    FSDR_INSUFFICIENT_INPUT  = 0,
+    FSDR_INSUFFICIENT_OUTPUT = 1,
+    FSDR_BOTH_SUFFICIENT     = 2,
+} fsdr_status;
+
+typedef struct {
+    size_t      consumed;
+    size_t      produced;
+    fsdr_status status;
+} fsdr_filter_result;
+
+/* Error codes (returned by every int-returning entry point). */
+enum {
+    FSDR_OK              = 0,
+    FSDR_ERR_NO_GPU      = 1,
+    FSDR_ERR_HIP         = 2,
+    FSDR_ERR_INVALID     = 3,
+    FSDR_ERR_UNSUPPORTED = 4,
+};
+
+/* ---- device management ---------------------------------------------- */
+int         fsdr_device_count(void);
+int         fsdr_set_device(int device);
+int         fsdr_synchronize(void);
+const char* fsdr_last_error(void);
+const char* fsdr_version(void);
+
+/* ---- Filter level ---------------------------------------------------- *
+ * Opaque stateless-filter handle (GPU-backed). Replaces the futuredsp
+ * cores behind the same span semantics. item types are fixed per
+ * constructor; *_host paths stage through persistent device buffers
+ * (PCIe-inclusive), *_dev paths take caller device pointers. */
+typedef struct fsdr_filter fsdr_filter;
+
+/* FirFilter<Complex32,Complex32,f32> — fir.rs:228-255 (f32 taps). */
+fsdr_filter* fsdr_fir_cf32_create(const float* taps, size_t n_taps);
+/* FirFilter<f32,f32,f32> — fir.rs:206-215. */
+fsdr_filter* fsdr_fir_f32_create(const float* taps, size_t n_taps);
+/* DecimatingFirFilter<Complex32,Complex32,f32> — decimating_fir.rs. */
+fsdr_filter* fsdr_decim_fir_cf32_create(size_t decimation,
+                                        const float* taps, size_t n_taps);
+/* PolyphaseResamplingFir<Complex32,Complex32,f32> — n_taps must be a
+ * multiple of interp (polyphase_resampling_fir.rs:54-56 assert). */
+fsdr_filter* fsdr_resamp_cf32_create(size_t interp, size_t decim,
+                                     const float* taps, size_t n_taps);
+/* Fft block — len must be a power of two in [16, 4096]; inverse/fft_shift
+ * flags and optional normalize factor as src/blocks/fft.rs:92-121. */
+fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse, int fft_shift,
+                                  const float* normalize);
+/* Apply |x|^2 (Complex32 -> f32) — the spectrum mag^2 map. */
+fsdr_filter* fsdr_mag2_create(void);
+
+/* Filter::length() — lib.rs:65-67. */
+size_t fsdr_filter_length(const fsdr_filter* f);
+
+/* filter() over host spans. Stages H2D/D2H internally. */
+int fsdr_filter_host(fsdr_filter* f, const void* in, size_t n_in,
+                     void* out, size_t n_out, fsdr_filter_result* r);
+/* filter() over device pointers, async on `stream` (no sync performed;
+ * the result math is computed on the host and returns immediately). */
+int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
+                    void* d_out, size_t n_out, void* stream,
+                    fsdr_filter_result* r);
+void fsdr_filter_destroy(fsdr_filter* f);
+
+/* Combine (2-input zip-map), complex-multiply variant. m = produced. */
+int fsdr_cmul_dev(const void* d_a, size_t n_a, const void* d_b, size_t n_b,
+                  void* d_out, size_t n_out, void* stream, size_t* m);
+int fsdr_cmul_host(const void* a, size_t n_a, const void* b, size_t n_b,
+                   void* out, size_t n_out, size_t* m);
+
+/* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
+int fsdr_dev_alloc(void** d_ptr, size_t bytes);
+int fsdr_dev_free(void* d_ptr);
+int fsdr_memcpy_h2d(void* d_dst, const void* src, size_t bytes);
+int fsdr_memcpy_d2h(void* dst, const void* d_src, size_t bytes);
+/* Fill a device buffer with n Complex32 samples, re/im iid uniform[-1,1)
+ * from a counter-based generator seeded by `seed` (deterministic,
+ * device-generated — the NullSource/bench synthetic source). */
+int fsdr_fill_uniform_cf32(void* d_ptr, size_t n, uint64_t seed,
+                           uint64_t offset, void* stream);
+
+/* ---- firdes (host-side tap designers; no GPU required) ---------------- *
+ * Mirror crates/futuredsp/src/firdes/basic.rs + windows.rs:144 +
+ * math/special_funs.rs:22-45 — the tap-generation row of SURVEY.md §8a. */
+double fsdr_kaiser_beta(double max_ripple); /* basic.rs:444-452 */
+void   fsdr_kaiser_window(size_t len, double beta, double* out);
+/* firdes::kaiser::lowpass<f32> (basic.rs:310-321). Returns tap count;
+ * fills out up to cap (query with out=NULL, cap=0). */
+size_t fsdr_firdes_kaiser_lowpass_f32(double cutoff, double transition_bw,
+                                      double max_ripple, float* out,
+                                      size_t cap);
+/* firdes::lowpass<f32> over a kaiser window of explicit length
+ * (basic.rs:25-42 with windows::kaiser) — fixed-length designer used by
+ * the bench (127 taps). */
+int fsdr_firdes_lowpass_kaiser_n_f32(size_t n_taps, double beta,
+                                     double cutoff, float* out);
+
+/* ---- Chain level ------------------------------------------------------ *
+ * Fused hot path: Fir(taps1) -> DecimatingFir(decim, taps2) -> Fft(len).
+ * Mirrors BASELINE configs[2]/[3]; per-stage math identical to the three
+ * filters composed (intermediates stay resident in HBM). */
+typedef struct fsdr_chain fsdr_chain;
+fsdr_chain* fsdr_chain_create(const float* taps1, size_t n_taps1,
+                              const float* taps2, size_t n_taps2,
+                              size_t decim, size_t fft_len);
+/* Run over device input; writes `frames*fft_len` Complex32 spectra to
+ * d_out (may be NULL with out_cap 0 -> spectra discarded into an internal
+ * buffer, NullSink-style). If d_mag is non-NULL also writes f32 |X|^2 of
+ * each bin (the config-4 spectrum join payload). Async on stream. */
+int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in, size_t n_in,
+                       void* d_out, size_t out_cap,
+                       void* d_mag, size_t mag_cap,
+                       void* stream, size_t* consumed, size_t* produced);
+void fsdr_chain_destroy(fsdr_chain* c);
+
+/* ---- Ring (Slab-style stream buffer) ---------------------------------- *
+ * N pinned-host buffers + N device mirrors circulating between an empty
+ * queue (writer side) and a full queue (reader side), with a
+ * reserved_items history prefix maintained exactly like slab.rs:369-399:
+ * the first `reserved` items of each acquired read-slice are the tail of
+ * the previous buffer. Writer: acquire -> fill host slice -> commit(n)
+ * (enqueues async H2D on the ring's copy stream). Reader: acquire (waits
+ * for the H2D event; yields the device pointer including history prefix)
+ * -> ... launch kernels ... -> release (recycles the buffer, copies the
+ * history tail). Single-producer single-consumer. */
+typedef struct fsdr_ring fsdr_ring;
+fsdr_ring* fsdr_ring_create(size_t n_buffers, size_t items_per_buffer,
+                            size_t item_bytes, size_t reserved_items);
+int  fsdr_ring_writer_acquire(fsdr_ring* r, void** host_ptr, size_t* items);
+int  fsdr_ring_writer_commit(fsdr_ring* r, size_t items);
+int  fsdr_ring_reader_acquire(fsdr_ring* r, void** dev_ptr, size_t* items);
+int  fsdr_ring_reader_release(fsdr_ring* r);
+void fsdr_ring_destroy(fsdr_ring* r);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* FUTURESDR_HIP_H */

@@ -1,0 +1,469 @@
+"""GPU parity tests: HIP kernels vs the KAT-pinned CPU oracle.
+
+All marked @pytest.mark.gpu — they run on a real MI355X (gpurun / driver).
+Everything compares through the C-ABI (the drop-in boundary); nothing here
+reads /root/reference at run time.
+
+Tolerance bar: consumed/produced/status bit-exact; float
+results |gpu - oracle|_inf <= 1e-5 * max(1, |oracle|_inf) for FIR-class
+kernels (same-order fp32 sums, GPU uses fma), 1e-4 relative l2 for FFT.
+"""
+import ctypes
+import os
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+GOLDEN = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden",
+                      "golden.npz")
+
+
+def rng(seed=0):
+    return np.random.default_rng(seed)
+
+
+def cplx(r, n):
+    return (r.uniform(-1, 1, (n, 2)) @ [1, 1j]).astype(np.complex64)
+
+
+def assert_close(got, ref, tol=1e-5):
+    ref = np.asarray(ref)
+    got = np.asarray(got)
+    assert got.shape == ref.shape
+    scale = max(1.0, float(np.abs(ref).max()) if ref.size else 1.0)
+    err = float(np.abs(got - ref).max()) if ref.size else 0.0
+    assert err <= tol * scale, f"max err {err} > {tol}*{scale}"
+
+
+# ---------------- FIR cf32 (the dominant kernel) ----------------------
+
+@pytest.mark.parametrize("n_in", [1, 126, 127, 128, 300, 1149, 1150, 1151,
+                                  2048, 4096, 12345, 100000])
+def test_fir_cf32_parity_sizes(gpu, oracle_lib, n_in):
+    r = rng(n_in)
+    taps = r.uniform(-1, 1, 127).astype(np.float32)
+    x = cplx(r, n_in)
+    f = gpu.Fir(taps)
+    assert f.length == 127
+    got, c, p, s = f.filter(x, n_in)
+    ref, co, po, so = oracle_lib.fir_cf32(taps, x, n_in)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref)
+
+
+@pytest.mark.parametrize("n_taps", [1, 2, 3, 5, 6, 7, 63, 64, 121, 127, 128,
+                                    250])
+def test_fir_cf32_parity_tap_counts(gpu, oracle_lib, n_taps):
+    r = rng(n_taps + 1000)
+    taps = r.uniform(-1, 1, n_taps).astype(np.float32)
+    x = cplx(r, 3000)
+    got, c, p, s = gpu.Fir(taps).filter(x, 4000)
+    ref, co, po, so = oracle_lib.fir_cf32(taps, x, 4000)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref)
+
+
+def test_fir_cf32_statuses(gpu, oracle_lib):
+    r = rng(7)
+    taps = r.uniform(-1, 1, 3).astype(np.float32)
+    x = cplx(r, 5)
+    f = gpu.Fir(taps)
+    # InsufficientOutput (fir.rs:71)
+    got, c, p, s = f.filter(x, 2)
+    assert (c, p, s) == (2, 2, gpu.INSUFFICIENT_OUTPUT)
+    # BothSufficient (fir.rs:72)
+    got, c, p, s = f.filter(x, 3)
+    assert (c, p, s) == (3, 3, gpu.BOTH_SUFFICIENT)
+    # InsufficientInput (fir.rs:73)
+    got, c, p, s = f.filter(x, 10)
+    assert (c, p, s) == (3, 3, gpu.INSUFFICIENT_INPUT)
+    # empty input / empty output
+    got, c, p, s = f.filter(x[:0], 10)
+    assert (c, p, s) == (0, 0, gpu.INSUFFICIENT_INPUT)
+    got, c, p, s = f.filter(x, 0)
+    assert (c, p, s) == (0, 0, gpu.INSUFFICIENT_OUTPUT)
+    # input shorter than taps
+    got, c, p, s = f.filter(x[:2], 10)
+    assert (c, p, s) == (0, 0, gpu.INSUFFICIENT_INPUT)
+
+
+def test_fir_cf32_impulse_exact(gpu):
+    """Impulse train -> each output is exactly one tap (bit-exact)."""
+    taps = rng(9).uniform(-1, 1, 127).astype(np.float32)
+    n = 8192
+    x = np.zeros(n, np.complex64)
+    x[::512] = 1.0 + 0.0j
+    got, c, p, s = gpu.Fir(taps).filter(x, n)
+    # y[k] = sum_t x[k+t] h[126-t]; impulses at k+t = 512*m
+    ref = np.zeros(p, np.complex64)
+    for m in range(0, n, 512):
+        for k in range(max(0, m - 126), min(p, m + 1)):
+            ref[k] += taps[126 - (m - k)]
+    np.testing.assert_array_equal(got, ref)
+
+
+def test_fir_cf32_vs_golden(gpu):
+    g = np.load(GOLDEN)
+    got, c, p, s = gpu.Fir(g["fir127_taps"]).filter(g["fir127_in"], 10 ** 6)
+    assert p == g["fir127_out"].size
+    assert_close(got, g["fir127_out"].astype(np.complex64))
+
+
+def test_fir_f32_parity(gpu, oracle_lib):
+    r = rng(11)
+    taps = r.uniform(-1, 1, 64).astype(np.float32)  # perf/fir shape
+    x = r.uniform(-1, 1, 15000).astype(np.float32)
+    got, c, p, s = gpu.FirF32(taps).filter(x, 15000)
+    ref, co, po, so = oracle_lib.fir_f32(taps, x, 15000)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref)
+
+
+# ---------------- Decimating FIR --------------------------------------
+
+@pytest.mark.parametrize("decim,n_taps,n_in", [
+    (4, 127, 50000),   # fast path, metric shape
+    (4, 127, 4222),    # one tile exactly
+    (4, 127, 4223), (4, 127, 9000), (4, 64, 10000), (4, 1, 1000),
+    (2, 127, 10000),   # generic path
+    (3, 33, 5000), (8, 127, 30000), (1, 127, 5000),
+])
+def test_decim_fir_parity(gpu, oracle_lib, decim, n_taps, n_in):
+    r = rng(decim * 1000 + n_taps)
+    taps = r.uniform(-1, 1, n_taps).astype(np.float32)
+    x = cplx(r, n_in)
+    got, c, p, s = gpu.DecimFir(decim, taps).filter(x, n_in)
+    ref, co, po, so = oracle_lib.decim_fir_cf32(decim, taps, x, n_in)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref)
+
+
+def test_decim_fir_statuses(gpu, oracle_lib):
+    r = rng(13)
+    taps = r.uniform(-1, 1, 3).astype(np.float32)
+    f = gpu.DecimFir(2, taps)
+    for n_in, n_out in [(4, 3), (5, 3), (5, 1), (6, 1), (6, 3), (6, 0),
+                        (0, 4), (2, 4)]:
+        x = cplx(r, n_in)
+        got, c, p, s = f.filter(x, n_out)
+        ref, co, po, so = oracle_lib.decim_fir_cf32(2, taps, x, n_out)
+        assert (c, p, s) == (co, po, so), (n_in, n_out)
+        assert_close(got, ref)
+
+
+# ---------------- Polyphase resampler ---------------------------------
+
+@pytest.mark.parametrize("interp,decim,n_taps", [
+    (3, 2, 6), (2, 1, 2), (1, 3, 2), (1, 4, 128), (5, 3, 40), (4, 7, 48),
+])
+def test_resampler_parity(gpu, oracle_lib, interp, decim, n_taps):
+    r = rng(interp * 100 + decim)
+    taps = r.uniform(-1, 1, n_taps).astype(np.float32)
+    x = cplx(r, 5000)
+    got, c, p, s = gpu.Resampler(interp, decim, taps).filter(x, 20000)
+    ref, co, po, so = oracle_lib.resamp_cf32(interp, decim, taps, x, 20000)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref)
+
+
+def test_resampler_output_multiple_of_interp(gpu, oracle_lib):
+    # InsufficientOutput rounds produced down to a multiple of interp
+    # (polyphase_resampling_fir.rs:96-100)
+    r = rng(17)
+    taps = r.uniform(-1, 1, 6).astype(np.float32)
+    x = cplx(r, 100)
+    got, c, p, s = gpu.Resampler(3, 2, taps).filter(x, 7)
+    ref, co, po, so = oracle_lib.resamp_cf32(3, 2, taps, x, 7)
+    assert (c, p, s) == (co, po, so)
+    assert p % 3 == 0
+    assert_close(got, ref)
+
+
+# ---------------- FFT ---------------------------------------------------
+
+@pytest.mark.parametrize("n", [16, 64, 256, 1024, 2048])
+def test_fft_parity_vs_oracle(gpu, oracle_lib, n):
+    r = rng(n)
+    frames = 8
+    x = cplx(r, n * frames)
+    got, c, p, s = gpu.Fft(n).filter(x, n * frames)
+    assert (c, p) == (n * frames, n * frames)
+    ref = np.concatenate([
+        oracle_lib.dft_cf32(x[i * n:(i + 1) * n]) for i in range(frames)])
+    rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+    assert rel < 1e-4, rel
+
+
+def test_fft_vs_golden(gpu):
+    g = np.load(GOLDEN)
+    for n in (64, 256, 1024):
+        xs, refs = g[f"fft{n}_in"], g[f"fft{n}_out"]
+        got, c, p, s = gpu.Fft(n).filter(xs.ravel(), xs.size)
+        got = got.reshape(xs.shape)
+        rel = np.linalg.norm(got - refs) / np.linalg.norm(refs)
+        assert rel < 1e-4, (n, rel)
+
+
+def test_fft_m_rounding_and_cap(gpu, oracle_lib):
+    r = rng(31)
+    x = cplx(r, 1024 * 40)
+    f = gpu.Fft(1024)
+    got, c, p, s = f.filter(x, 1024 * 40)
+    assert c == p == 1024 * 32  # BUFF_FFTS cap (fft.rs:56,171)
+    got, c, p, s = f.filter(x[:2500], 4096)
+    assert c == p == 2048  # rounded down to multiple of len
+    got, c, p, s = f.filter(x[:1000], 4096)
+    assert c == p == 0
+
+
+def test_fft_inverse_shift_normalize(gpu, oracle_lib):
+    r = rng(37)
+    n = 256
+    x = cplx(r, n * 2)
+    for kw in [dict(inverse=True), dict(fft_shift=True),
+               dict(inverse=True, fft_shift=True),
+               dict(normalize=1.0 / n), dict(fft_shift=True,
+                                             normalize=0.5)]:
+        got, c, p, s = gpu.Fft(n, **kw).filter(x, n * 2)
+        ref, m = oracle_lib.fft_block(n, x, n * 2, **kw)
+        assert p == m
+        rel = np.linalg.norm(got - ref) / max(np.linalg.norm(ref), 1e-30)
+        assert rel < 1e-4, (kw, rel)
+
+
+def test_fft_roundtrip(gpu):
+    r = rng(41)
+    n = 1024
+    x = cplx(r, n * 4)
+    X, _, _, _ = gpu.Fft(n).filter(x, x.size)
+    back, _, _, _ = gpu.Fft(n, inverse=True, normalize=1.0 / n).filter(
+        X, X.size)
+    assert_close(back, x, 1e-4)
+
+
+def test_fft_parseval(gpu):
+    r = rng(43)
+    n = 1024
+    x = cplx(r, n * 16)
+    X, _, _, _ = gpu.Fft(n).filter(x, x.size)
+    for f in range(16):
+        e_t = np.sum(np.abs(x[f * n:(f + 1) * n]) ** 2)
+        e_f = np.sum(np.abs(X[f * n:(f + 1) * n]) ** 2) / n
+        assert abs(e_t - e_f) / e_t < 1e-5
+
+
+# ---------------- element-wise + synthetic source ----------------------
+
+def test_mag2_parity(gpu, oracle_lib):
+    x = cplx(rng(47), 10000)
+    got, c, p, s = gpu.Mag2().filter(x, 10000)
+    assert_close(got, oracle_lib.mag2(x), 1e-6)
+
+
+def test_cmul_parity(gpu, oracle_lib):
+    r = rng(53)
+    a, b = cplx(r, 5000), cplx(r, 4000)
+    got = gpu.cmul_host(a, b)
+    ref = oracle_lib.cmul(a, b)
+    assert got.size == ref.size == 4000
+    assert_close(got, ref, 1e-6)
+
+
+def test_fill_uniform_deterministic(gpu):
+    lib = gpu.lib()
+    n = 100000
+    ptr = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(ptr), n * 8) == 0
+    try:
+        gpu.fill_uniform_dev(ptr.value, n, seed=123)
+        h1 = np.zeros(n, np.complex64)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(h1.ctypes.data), ptr, n * 8)
+        gpu.fill_uniform_dev(ptr.value, n, seed=123)
+        h2 = np.zeros(n, np.complex64)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(h2.ctypes.data), ptr, n * 8)
+        np.testing.assert_array_equal(h1, h2)
+        v = h1.view(np.float32)
+        assert v.min() >= -1.0 and v.max() < 1.0
+        assert abs(v.mean()) < 0.01 and abs(v.std() - 0.577) < 0.01
+        # different seed differs
+        gpu.fill_uniform_dev(ptr.value, n, seed=124)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(h2.ctypes.data), ptr, n * 8)
+        assert not np.array_equal(h1, h2)
+    finally:
+        lib.fsdr_dev_free(ptr)
+
+
+# ---------------- chain (the bench hot path) ---------------------------
+
+def test_chain_parity_vs_oracle(gpu, oracle_lib):
+    r = rng(59)
+    n_in = 4 * 1024 * 8 + 4 * 127 + 500  # ~8 frames + slack
+    x = cplx(r, n_in)
+    taps1 = r.uniform(-1, 1, 127).astype(np.float32)
+    taps2 = r.uniform(-1, 1, 127).astype(np.float32)
+    lib = gpu.lib()
+    d_in = ctypes.c_void_p()
+    d_out = ctypes.c_void_p()
+    d_mag = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_in), n_in * 8) == 0
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), n_in * 8) == 0
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_mag), n_in * 4) == 0
+    try:
+        lib.fsdr_memcpy_h2d(d_in, ctypes.c_void_p(x.ctypes.data), n_in * 8)
+        ch = gpu.Chain(taps1, taps2, 4, 1024)
+        cons, prod = ch.run_dev(d_in.value, n_in, d_out.value, n_in,
+                                d_mag.value, n_in)
+        gpu.synchronize()
+        ref, cons_ref = oracle_lib.chain_cf32(taps1, taps2, 4, 1024, x)
+        assert prod == ref.size and cons == cons_ref
+        got = np.zeros(prod, np.complex64)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(got.ctypes.data), d_out,
+                            prod * 8)
+        rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+        assert rel < 1e-4, rel
+        mag = np.zeros(prod, np.float32)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(mag.ctypes.data), d_mag,
+                            prod * 4)
+        assert_close(mag, np.abs(got) ** 2, 1e-6)
+    finally:
+        lib.fsdr_dev_free(d_in)
+        lib.fsdr_dev_free(d_out)
+        lib.fsdr_dev_free(d_mag)
+
+
+def test_chain_full_size_properties(gpu):
+    """BASELINE-scale run (2^22 samples): determinism (checksum of two
+    runs identical) + per-frame Parseval between stage-2 output computed
+    through the filter ABI and the chain's spectra."""
+    r = rng(61)
+    lib = gpu.lib()
+    n_in = 1 << 22
+    taps1 = gpu.kaiser_lowpass(0.1, 0.02, 1e-4)
+    taps2 = gpu.lowpass_kaiser_n(127, gpu.kaiser_beta(1e-4), 0.11)
+    d_in = ctypes.c_void_p()
+    d_out = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_in), n_in * 8) == 0
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), n_in * 8) == 0
+    try:
+        gpu.fill_uniform_dev(d_in.value, n_in, seed=0x5D5D5D5D)
+        ch = gpu.Chain(taps1, taps2, 4, 1024)
+        cons, prod = ch.run_dev(d_in.value, n_in, d_out.value, n_in)
+        gpu.synchronize()
+        assert prod > 0
+        out1 = np.zeros(prod, np.complex64)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(out1.ctypes.data), d_out,
+                            prod * 8)
+        # determinism
+        cons2, prod2 = ch.run_dev(d_in.value, n_in, d_out.value, n_in)
+        gpu.synchronize()
+        out2 = np.zeros(prod, np.complex64)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(out2.ctypes.data), d_out,
+                            prod * 8)
+        assert (cons, prod) == (cons2, prod2)
+        np.testing.assert_array_equal(out1, out2)
+        # Parseval vs the composed per-stage path at full size
+        fir1 = gpu.Fir(taps1)
+        d_y1 = ctypes.c_void_p()
+        assert lib.fsdr_dev_alloc(ctypes.byref(d_y1), n_in * 8) == 0
+        try:
+            c1, p1, _ = fir1.filter_dev(d_in.value, n_in, d_y1.value, n_in)
+            fir2 = gpu.DecimFir(4, taps2)
+            d_y2 = ctypes.c_void_p()
+            assert lib.fsdr_dev_alloc(ctypes.byref(d_y2), n_in * 2) == 0
+            try:
+                c2, p2, _ = fir2.filter_dev(d_y1.value, p1, d_y2.value,
+                                            n_in // 4)
+                gpu.synchronize()
+                y2 = np.zeros(p2, np.complex64)
+                lib.fsdr_memcpy_d2h(ctypes.c_void_p(y2.ctypes.data), d_y2,
+                                    p2 * 8)
+                frames = prod // 1024
+                for f in range(0, frames, max(1, frames // 7)):
+                    e_t = np.sum(np.abs(y2[f * 1024:(f + 1) * 1024]) ** 2)
+                    e_f = np.sum(np.abs(out1[f * 1024:(f + 1) * 1024]) ** 2)
+                    assert abs(e_f / 1024 - e_t) / max(e_t, 1e-30) < 1e-4
+            finally:
+                lib.fsdr_dev_free(d_y2)
+        finally:
+            lib.fsdr_dev_free(d_y1)
+    finally:
+        lib.fsdr_dev_free(d_in)
+        lib.fsdr_dev_free(d_out)
+
+
+# ---------------- ring (Slab-style) ------------------------------------
+
+def test_ring_history_prefix(gpu, oracle_lib):
+    """Stream a long signal through the pinned ring in chunks and run the
+    FIR on each acquired device buffer; the reserved prefix must carry
+    exactly taps-1 samples of history so the concatenated outputs equal
+    the single-shot oracle (slab.rs reserved-prefix semantics)."""
+    lib = gpu.lib()
+    r = rng(67)
+    taps = r.uniform(-1, 1, 127).astype(np.float32)
+    n_total, chunk = 40960, 4096
+    x = cplx(r, n_total)
+    reserved = 126
+    ring = lib.fsdr_ring_create(4, chunk, 8, reserved)
+    assert ring
+    try:
+        fir = gpu.Fir(taps)
+        d_out = ctypes.c_void_p()
+        assert lib.fsdr_dev_alloc(ctypes.byref(d_out), chunk * 8) == 0
+        outs = []
+        try:
+            for off in range(0, n_total, chunk):
+                hp = ctypes.c_void_p()
+                items = ctypes.c_size_t()
+                assert lib.fsdr_ring_writer_acquire(
+                    ring, ctypes.byref(hp), ctypes.byref(items)) == 0
+                n = min(chunk, n_total - off)
+                ctypes.memmove(hp, ctypes.c_void_p(
+                    x[off:off + n].ctypes.data), n * 8)
+                assert lib.fsdr_ring_writer_commit(ring, n) == 0
+                dp = ctypes.c_void_p()
+                got_items = ctypes.c_size_t()
+                assert lib.fsdr_ring_reader_acquire(
+                    ring, ctypes.byref(dp), ctypes.byref(got_items)) == 0
+                n_in = reserved + got_items.value
+                c, p, s = fir.filter_dev(dp.value, n_in, d_out.value, chunk)
+                gpu.synchronize()
+                h = np.zeros(p, np.complex64)
+                lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
+                                    p * 8)
+                outs.append(h)
+                assert lib.fsdr_ring_reader_release(ring) == 0
+        finally:
+            lib.fsdr_dev_free(d_out)
+        got = np.concatenate(outs)
+        # first chunk sees a zero history prefix; the oracle equivalent is
+        # the signal with 126 zeros prepended
+        xz = np.concatenate([np.zeros(reserved, np.complex64), x])
+        ref, c, p, s = oracle_lib.fir_cf32(taps, xz, got.size)
+        assert p == got.size
+        assert_close(got, ref)
+    finally:
+        lib.fsdr_ring_destroy(ring)
+
+
+# ---------------- dev path on a torch stream ---------------------------
+
+def test_filter_dev_on_torch_stream(gpu):
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.fail("torch.cuda not available on GPU box")
+    r = rng(71)
+    taps = r.uniform(-1, 1, 127).astype(np.float32)
+    x = torch.from_numpy(cplx(r, 50000).view(np.float32)).cuda().view(-1, 2)
+    y = torch.empty_like(x)
+    f = gpu.Fir(taps)
+    st = torch.cuda.current_stream().cuda_stream
+    c, p, s = f.filter_dev(x.data_ptr(), x.shape[0], y.data_ptr(),
+                           y.shape[0], stream=st)
+    torch.cuda.synchronize()
+    got = y[:p].cpu().numpy().view(np.complex64).ravel()
+    ref, _, _, _ = __import__("oracle").fir_cf32(
+        taps, x.cpu().numpy().view(np.complex64).ravel(), y.shape[0])
+    assert_close(got, ref)
