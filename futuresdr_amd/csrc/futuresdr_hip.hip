@@ -605,7 +605,7 @@ static int ensure_dev(void** p, size_t* cur, size_t want) {
 static int upload_taps_padded(fsdr_filter* f, const float* taps, size_t nt,
                               int mod) {
     size_t tp = nt;
-    while (tp % mod != 1) tp++;
+    while (tp % mod != 1 % mod) tp++;
     std::vector<float> h(tp, 0.f);
     memcpy(h.data() + (tp - nt), taps, nt * sizeof(float));
     HIP_TRY(hipMalloc(&f->d_taps, tp * sizeof(float)));
