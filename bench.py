@@ -21,6 +21,11 @@ import time
 
 import numpy as np
 
+# Import torch (and its bundled HIP runtime) BEFORE the futuresdr_amd C-ABI
+# library: loading /opt/rocm's libamdhip64 first makes torch's lazy CUDA/HIP
+# init fail with "No HIP GPUs are available" (observed on the GPU box).
+import torch  # noqa: E402
+
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
@@ -141,14 +146,15 @@ def main():
     local = int(os.environ.get("LOCAL_RANK", str(rank)))
     n_gpus = max(args.gpus, world)
 
+    if not torch.cuda.is_available():
+        raise SystemExit("bench.py needs a HIP device (no CPU fallback)")
+    torch.cuda.set_device(local)
+
     import futuresdr_amd as fa
     if fa.device_count() < 1:
-        raise SystemExit("bench.py needs a HIP device (no CPU fallback)")
+        raise SystemExit("futuresdr_amd sees no HIP device")
     fa.set_device(local)
     lib = fa.lib()
-
-    import torch
-    torch.cuda.set_device(local)
     st = torch.cuda.current_stream()
 
     td = None
