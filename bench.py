@@ -115,27 +115,28 @@ def measure_cpu_baseline(taps1, taps2, decim, fft_len, fixed_sample):
     cores = os.cpu_count() or 1
     rng = np.random.default_rng(0x5D5D5D5D)
 
-    def run(n):
-        x = (rng.uniform(-1, 1, (n, 2)) @ [1, 1j]).astype(np.complex64)
-        t0 = time.perf_counter()
-        _, consumed = oracle.chain_cf32(taps1, taps2, decim, fft_len, x,
-                                        capture=False, nthreads=0)
+    n = fixed_sample or (1 << 26)
+    x = (rng.uniform(-1, 1, (n, 2)) @ [1, 1j]).astype(np.complex64)
+    # repeat passes over the same buffer until >= ~10 s of CPU work
+    consumed = 0
+    passes = 0
+    t0 = time.perf_counter()
+    while True:
+        _, c = oracle.chain_cf32(taps1, taps2, decim, fft_len, x,
+                                 capture=False, nthreads=0)
+        consumed += c
+        passes += 1
         dt = time.perf_counter() - t0
-        return consumed, dt
-
-    n = fixed_sample or (1 << 22)
-    consumed, dt = run(n)
-    if not fixed_sample and dt < 4.0:
-        n = min(1 << 27, max(n * 2, int(n * 8.0 / max(dt, 1e-3))))
-        consumed, dt = run(n)
+        if dt >= 10.0 or (fixed_sample and passes >= 1) or passes >= 64:
+            break
     msps = consumed / dt / 1e6
     return {
         "value": round(msps, 2),
         "unit": "MSample/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"{n} Complex32 samples through the oracle chain, "
-                  f"OpenMP {cores} threads, {dt:.1f}s",
+        "sample": f"{passes} pass(es) over {n} Complex32 samples through "
+                  f"the oracle chain, OpenMP {cores} threads, {dt:.1f}s",
     }
 
 
