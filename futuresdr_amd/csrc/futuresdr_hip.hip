@@ -118,11 +118,14 @@ __device__ __forceinline__ float2 cmulf(float2 a, float2 b) {
  * h[T-1-t], accumulating re/im separately in fp32.
  *
  * Tiling: 256 lanes/block, R=4 consecutive outputs per lane, input tile in
- * SoA LDS planes, 3-pair rotating register window (one ds_read_b64 per
- * plane per 2-tap step feeds 16 FMAs) -> VALU-bound with LDS at ~50%.
- * Requires n_taps_padded % 6 == 1 (host pads taps with leading zeros in
- * stored order = trailing zeros in reversed application order; padded taps
- * multiply staged zeros only). */
+ * SoA LDS planes (pad-2-per-16 keeps the 16 B lane-stride pair reads
+ * conflict-free). Inner loop: 2 taps per step, 16 FMAs per step; the
+ * window pair AND the (LDS-staged, reversed) tap pair for step s+1 are
+ * prefetched during step s so no FMA waits on a just-issued LDS read
+ * (the v1 kernel was SQ_WAIT_INST-bound at 28% of fp32 peak for exactly
+ * that reason — profiles/rocprof_r01*). Requires n_taps_padded % 8 == 1
+ * (host pads taps with leading zeros; padded taps multiply staged zeros
+ * only). */
 
 #define FIR_BLOCK 256
 #define FIR_R 4
@@ -132,73 +135,79 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32(
     const float2* __restrict__ in, float2* __restrict__ out,
     const float* __restrict__ taps, int n_taps_padded, long long n_out,
     long long n_in_valid) {
-    const int tp = n_taps_padded;                 /* tp % 6 == 1 */
-    const unsigned elems = FIR_TILE_OUT + tp - 1; /* staged per tile */
+    const int tp = n_taps_padded;                     /* tp % 8 == 1 */
+    const unsigned elems = FIR_TILE_OUT + tp - 1 + 6; /* + prefetch slack */
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* s_re = (float*)smem;
     float* s_im = s_re + plane_floats(elems);
+    float* s_rt = s_im + plane_floats(elems); /* reversed taps, tp+1 slots */
 
     const int tid = threadIdx.x;
     for (long long tile = blockIdx.x;
          tile * (long long)FIR_TILE_OUT < n_out; tile += gridDim.x) {
         const long long out_base = tile * FIR_TILE_OUT;
-        /* stage input tile (zero-fill beyond valid input) */
         for (unsigned i = tid; i < elems; i += FIR_BLOCK) {
             long long g = out_base + i;
             float2 v = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
             s_re[lds_pad(i)] = v.x;
             s_im[lds_pad(i)] = v.y;
         }
+        for (int i = tid; i <= tp; i += FIR_BLOCK)
+            s_rt[i] = (i < tp) ? taps[tp - 1 - i] : 0.f;
         __syncthreads();
 
         const unsigned eb = (unsigned)tid * FIR_R; /* tile-relative base */
         float ar[FIR_R] = {0.f, 0.f, 0.f, 0.f};
         float ai[FIR_R] = {0.f, 0.f, 0.f, 0.f};
-        float wre[6], wim[6];
-        /* preload window elements eb..eb+3 (two aligned pairs) */
-        {
-            float2 p0 = *(const float2*)&s_re[lds_pad(eb)];
-            float2 p1 = *(const float2*)&s_re[lds_pad(eb + 2)];
-            wre[0] = p0.x; wre[1] = p0.y; wre[2] = p1.x; wre[3] = p1.y;
-            float2 q0 = *(const float2*)&s_im[lds_pad(eb)];
-            float2 q1 = *(const float2*)&s_im[lds_pad(eb + 2)];
-            wim[0] = q0.x; wim[1] = q0.y; wim[2] = q1.x; wim[3] = q1.y;
+        /* 4-pair rotating window: slot q holds pair q mod 4 (elements
+         * 2q, 2q+1 relative to eb), flat layout w[2*slot + parity]. */
+        float wre[8], wim[8];
+#pragma unroll
+        for (int p = 0; p < 3; p++) {
+            float2 pr = *(const float2*)&s_re[lds_pad(eb + 2 * p)];
+            float2 pi = *(const float2*)&s_im[lds_pad(eb + 2 * p)];
+            wre[2 * p] = pr.x; wre[2 * p + 1] = pr.y;
+            wim[2 * p] = pi.x; wim[2 * p + 1] = pi.y;
         }
-        const int pair_steps = (tp - 1) / 2; /* multiple of 3 */
+        float2 ht0 = *(const float2*)&s_rt[0];
+        const int pair_steps = (tp - 1) / 2; /* multiple of 4 */
         int s = 0;
-        /* one 2-tap step at unroll phase U (s % 3 == U): window flat index
-         * of tile-relative element e is (e - eb) % 6 */
+        /* flat window index of tile-relative element e = 2s + d at phase
+         * U = s%4: 2*((U + d/2) % 4) + d%2, d in [0,5] */
+#define FIR_W(U, d) (2 * (((U) + (d) / 2) % 4) + (d) % 2)
 #define FIR_STEP(U)                                                          \
     do {                                                                     \
-        {                                                                    \
-            float2 nr = *(const float2*)&s_re[lds_pad(eb + 2 * s + 4)];      \
-            float2 ni = *(const float2*)&s_im[lds_pad(eb + 2 * s + 4)];      \
-            wre[(2 * (U) + 4) % 6] = nr.x; wre[(2 * (U) + 5) % 6] = nr.y;    \
-            wim[(2 * (U) + 4) % 6] = ni.x; wim[(2 * (U) + 5) % 6] = ni.y;    \
-        }                                                                    \
-        const float h0 = taps[tp - 1 - 2 * s];                               \
-        const float h1 = taps[tp - 2 - 2 * s];                               \
+        float2 nr = *(const float2*)&s_re[lds_pad(eb + 2 * s + 6)];          \
+        float2 ni = *(const float2*)&s_im[lds_pad(eb + 2 * s + 6)];          \
+        float2 ht1 = *(const float2*)&s_rt[2 * s + 2];                       \
         _Pragma("unroll") for (int j = 0; j < FIR_R; j++) {                  \
-            ar[j] = fmaf(wre[(2 * (U) + j) % 6], h0, ar[j]);                 \
-            ai[j] = fmaf(wim[(2 * (U) + j) % 6], h0, ai[j]);                 \
-            ar[j] = fmaf(wre[(2 * (U) + j + 1) % 6], h1, ar[j]);             \
-            ai[j] = fmaf(wim[(2 * (U) + j + 1) % 6], h1, ai[j]);             \
+            ar[j] = fmaf(wre[FIR_W(U, j)], ht0.x, ar[j]);                    \
+            ai[j] = fmaf(wim[FIR_W(U, j)], ht0.x, ai[j]);                    \
+            ar[j] = fmaf(wre[FIR_W(U, j + 1)], ht0.y, ar[j]);                \
+            ai[j] = fmaf(wim[FIR_W(U, j + 1)], ht0.y, ai[j]);                \
         }                                                                    \
+        wre[2 * (((U) + 3) % 4)] = nr.x;                                     \
+        wre[2 * (((U) + 3) % 4) + 1] = nr.y;                                 \
+        wim[2 * (((U) + 3) % 4)] = ni.x;                                     \
+        wim[2 * (((U) + 3) % 4) + 1] = ni.y;                                 \
+        ht0 = ht1;                                                           \
         s++;                                                                 \
     } while (0)
         for (; s < pair_steps;) {
             FIR_STEP(0);
             FIR_STEP(1);
             FIR_STEP(2);
+            FIR_STEP(3);
         }
 #undef FIR_STEP
-        /* final (unpaired) tap t = tp-1: flat index (tp-1+j) % 6 == j */
+        /* final (unpaired) tap t = tp-1: element e = tp-1+j is in pair
+         * ps + j/2 -> slot j/2 (ps % 4 == 0), parity j%2 */
         {
-            const float h0 = taps[0];
+            const float h0 = s_rt[tp - 1];
 #pragma unroll
             for (int j = 0; j < FIR_R; j++) {
-                ar[j] = fmaf(wre[j], h0, ar[j]);
-                ai[j] = fmaf(wim[j], h0, ai[j]);
+                ar[j] = fmaf(wre[2 * (j / 2) + (j % 2)], h0, ar[j]);
+                ai[j] = fmaf(wim[2 * (j / 2) + (j % 2)], h0, ai[j]);
             }
         }
 #pragma unroll
@@ -214,10 +223,11 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32(
  * Restates decimating_fir.rs:80-95 for Complex<f32>/f32: y[k] =
  * sum_t x[D-1 + k*D + t] * h[T-1-t].
  *
- * 256 lanes, R2=4 consecutive decimated outputs per lane (input span 16B
- * lane stride * 4 = 64 B -> pad keeps pair reads conflict-free), 7-pair
- * rotating window; taps processed as tap 0 (prologue, b32 reads) + pairs
- * (1,2),(3,4),...  Requires n_taps_padded % 14 == 1. */
+ * 256 lanes, R2=4 consecutive decimated outputs per lane; 8-pair rotating
+ * window with one-step-ahead prefetch of both the window pair and the
+ * LDS-staged reversed tap pair (same latency-hiding structure as
+ * k_fir_cf32). Taps processed as tap 0 (prologue) + pairs (1,2),(3,4),...
+ * Requires n_taps_padded % 16 == 1. */
 
 #define DFIR_BLOCK 256
 #define DFIR_R 4
@@ -228,11 +238,12 @@ __global__ __launch_bounds__(DFIR_BLOCK) void k_fir_decim4_cf32(
     const float* __restrict__ taps, int n_taps_padded, long long n_out,
     long long n_in_valid) {
     constexpr int D = 4;
-    const int tp = n_taps_padded; /* tp % 14 == 1 */
-    const unsigned elems = DFIR_TILE_OUT * D + tp - 1;
+    const int tp = n_taps_padded; /* tp % 16 == 1 */
+    const unsigned elems = DFIR_TILE_OUT * D + tp - 1 + 20;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* s_re = (float*)smem;
     float* s_im = s_re + plane_floats(elems);
+    float* s_rt = s_im + plane_floats(elems); /* reversed taps MINUS tap0 */
 
     const int tid = threadIdx.x;
     for (long long tile = blockIdx.x;
@@ -245,6 +256,9 @@ __global__ __launch_bounds__(DFIR_BLOCK) void k_fir_decim4_cf32(
             s_re[lds_pad(i)] = v.x;
             s_im[lds_pad(i)] = v.y;
         }
+        /* s_rt[i] = coefficient of tap t=i+1, i.e. taps[tp-2-i] */
+        for (int i = tid; i <= tp; i += DFIR_BLOCK)
+            s_rt[i] = (i < tp - 1) ? taps[tp - 2 - i] : 0.f;
         __syncthreads();
 
         /* lane-relative input element for (output j, tap t):
@@ -261,43 +275,42 @@ __global__ __launch_bounds__(DFIR_BLOCK) void k_fir_decim4_cf32(
                 ai[j] = fmaf(s_im[lds_pad(eb + 3 + 4 * j)], h, ai[j]);
             }
         }
-        /* window pairs p (pair p = elements 2p, 2p+1 relative to eb):
-         * step s covers taps (1+2s, 2+2s); output j reads pair s+2+2j.
-         * Preload pairs 2..7 into slots p%7; step s reads pair s+8. */
-        float wre[14], wim[14];
+        /* step s covers taps (1+2s, 2+2s); output j reads BOTH elements of
+         * pair s+2+2j; window = 8 rotating pair slots (p % 8), prefetch
+         * pair s+9 during step s. */
+        float wre[16], wim[16];
 #pragma unroll
-        for (int p = 2; p <= 7; p++) {
+        for (int p = 2; p <= 8; p++) {
             float2 pr = *(const float2*)&s_re[lds_pad(eb + 2 * p)];
             float2 pi = *(const float2*)&s_im[lds_pad(eb + 2 * p)];
-            wre[(p % 7) * 2] = pr.x; wre[(p % 7) * 2 + 1] = pr.y;
-            wim[(p % 7) * 2] = pi.x; wim[(p % 7) * 2 + 1] = pi.y;
+            wre[(p % 8) * 2] = pr.x; wre[(p % 8) * 2 + 1] = pr.y;
+            wim[(p % 8) * 2] = pi.x; wim[(p % 8) * 2 + 1] = pi.y;
         }
-        const int pair_steps = (tp - 1) / 2; /* multiple of 7 */
+        float2 ht0 = *(const float2*)&s_rt[0];
+        const int pair_steps = (tp - 1) / 2; /* multiple of 8 */
         int s = 0;
 #define DFIR_STEP(U)                                                         \
     do {                                                                     \
-        {                                                                    \
-            float2 nr = *(const float2*)&s_re[lds_pad(eb + 2 * (s + 8))];    \
-            float2 ni = *(const float2*)&s_im[lds_pad(eb + 2 * (s + 8))];    \
-            wre[(((U) + 8) % 7) * 2] = nr.x;                                 \
-            wre[(((U) + 8) % 7) * 2 + 1] = nr.y;                             \
-            wim[(((U) + 8) % 7) * 2] = ni.x;                                 \
-            wim[(((U) + 8) % 7) * 2 + 1] = ni.y;                             \
-        }                                                                    \
-        const float h0 = taps[tp - 2 - 2 * s];                               \
-        const float h1 = taps[tp - 3 - 2 * s];                               \
+        float2 nr = *(const float2*)&s_re[lds_pad(eb + 2 * (s + 9))];        \
+        float2 ni = *(const float2*)&s_im[lds_pad(eb + 2 * (s + 9))];        \
+        float2 ht1 = *(const float2*)&s_rt[2 * s + 2];                       \
         _Pragma("unroll") for (int j = 0; j < DFIR_R; j++) {                 \
-            const int sl = (((U) + 2 + 2 * j) % 7) * 2;                      \
-            ar[j] = fmaf(wre[sl], h0, ar[j]);                                \
-            ai[j] = fmaf(wim[sl], h0, ai[j]);                                \
-            ar[j] = fmaf(wre[sl + 1], h1, ar[j]);                            \
-            ai[j] = fmaf(wim[sl + 1], h1, ai[j]);                            \
+            const int sl = (((U) + 2 + 2 * j) % 8) * 2;                      \
+            ar[j] = fmaf(wre[sl], ht0.x, ar[j]);                             \
+            ai[j] = fmaf(wim[sl], ht0.x, ai[j]);                             \
+            ar[j] = fmaf(wre[sl + 1], ht0.y, ar[j]);                         \
+            ai[j] = fmaf(wim[sl + 1], ht0.y, ai[j]);                         \
         }                                                                    \
+        wre[(((U) + 1) % 8) * 2] = nr.x;                                     \
+        wre[(((U) + 1) % 8) * 2 + 1] = nr.y;                                 \
+        wim[(((U) + 1) % 8) * 2] = ni.x;                                     \
+        wim[(((U) + 1) % 8) * 2 + 1] = ni.y;                                 \
+        ht0 = ht1;                                                           \
         s++;                                                                 \
     } while (0)
         for (; s < pair_steps;) {
             DFIR_STEP(0); DFIR_STEP(1); DFIR_STEP(2); DFIR_STEP(3);
-            DFIR_STEP(4); DFIR_STEP(5); DFIR_STEP(6);
+            DFIR_STEP(4); DFIR_STEP(5); DFIR_STEP(6); DFIR_STEP(7);
         }
 #undef DFIR_STEP
 #pragma unroll
@@ -631,7 +644,7 @@ extern "C" fsdr_filter* fsdr_fir_cf32_create(const float* taps,
     fsdr_filter* f = create_common(K_FIR_CF32);
     if (!f) return nullptr;
     f->n_taps = n_taps;
-    if (upload_taps_padded(f, taps, n_taps, 6) != FSDR_OK) {
+    if (upload_taps_padded(f, taps, n_taps, 8) != FSDR_OK) {
         delete f;
         return nullptr;
     }
@@ -668,7 +681,7 @@ extern "C" fsdr_filter* fsdr_decim_fir_cf32_create(size_t decimation,
     if (!f) return nullptr;
     f->n_taps = n_taps;
     f->decim = decimation;
-    int mod = (decimation == 4) ? 14 : 1; /* fast path needs tp%14==1 */
+    int mod = (decimation == 4) ? 16 : 1; /* fast path needs tp%14==1 */
     if (upload_taps_padded(f, taps, n_taps, mod) != FSDR_OK) {
         delete f;
         return nullptr;
@@ -764,8 +777,9 @@ static int grid_for(long long work_items, int block) {
 static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
                            size_t n_out, size_t n_in, hipStream_t st) {
     if (n_out == 0) return FSDR_OK;
-    unsigned elems = FIR_TILE_OUT + f->n_taps_padded - 1;
-    size_t lds = 2 * (size_t)plane_floats(elems) * sizeof(float);
+    unsigned elems = FIR_TILE_OUT + f->n_taps_padded - 1 + 6;
+    size_t lds = (2 * (size_t)plane_floats(elems) + f->n_taps_padded + 4) *
+                 sizeof(float);
     long long tiles = ((long long)n_out + FIR_TILE_OUT - 1) / FIR_TILE_OUT;
     int grid = (int)std::min<long long>(tiles, 256 * 16);
     hipLaunchKernelGGL(k_fir_cf32, dim3(grid), dim3(FIR_BLOCK), lds, st,
@@ -779,8 +793,9 @@ static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
                              size_t n_out, size_t n_in, hipStream_t st) {
     if (n_out == 0) return FSDR_OK;
     if (f->decim == 4) {
-        unsigned elems = DFIR_TILE_OUT * 4 + f->n_taps_padded - 1;
-        size_t lds = 2 * (size_t)plane_floats(elems) * sizeof(float);
+        unsigned elems = DFIR_TILE_OUT * 4 + f->n_taps_padded - 1 + 20;
+        size_t lds = (2 * (size_t)plane_floats(elems) + f->n_taps_padded + 4)
+                     * sizeof(float);
         long long tiles =
             ((long long)n_out + DFIR_TILE_OUT - 1) / DFIR_TILE_OUT;
         int grid = (int)std::min<long long>(tiles, 256 * 8);
