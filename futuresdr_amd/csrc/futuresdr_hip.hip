@@ -219,6 +219,92 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32(
     }
 }
 
+/* ---- Fully-unrolled compile-time-tap-count FIR variant --------------- *
+ * Same math as k_fir_cf32; TP is the padded tap count (TP % 4 == 1) and
+ * the device tap array is REVERSED (rt[i] = h[n_taps-1-i], zero-filled to
+ * TP). Linear SoA LDS planes (no pad): the window is read as float4
+ * groups at 16 B lane stride (the conflict-free ds_read_b128 pattern),
+ * addresses are affine in the unrolled loop (fold to immediate offsets),
+ * and full unroll renames the 2-group register window (no rotation movs
+ * -- the v2 kernel spent ~2/3 of its VALU on movs + padded-address math,
+ * profiles/rocprof_r01*). Taps come from global memory with constant
+ * offsets (scalar-cache loads, keeps LDS bandwidth for the window). */
+template <int TP>
+__global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtaps, long long n_out, long long n_in_valid) {
+    static_assert(TP % 4 == 1, "TP must be 1 mod 4");
+    const unsigned elems = FIR_TILE_OUT + TP + 8;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* s_re = (float*)smem;
+    float* s_im = s_re + ((elems + 7u) & ~7u);
+
+    const int tid = threadIdx.x;
+    for (long long tile = blockIdx.x;
+         tile * (long long)FIR_TILE_OUT < n_out; tile += gridDim.x) {
+        const long long out_base = tile * FIR_TILE_OUT;
+        for (unsigned i = tid; i < elems; i += FIR_BLOCK) {
+            long long g = out_base + i;
+            float2 v = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+            s_re[i] = v.x;
+            s_im[i] = v.y;
+        }
+        __syncthreads();
+
+        const unsigned eb = (unsigned)tid * FIR_R;
+        float ar[FIR_R] = {0.f, 0.f, 0.f, 0.f};
+        float ai[FIR_R] = {0.f, 0.f, 0.f, 0.f};
+        float4 r0 = *(const float4*)&s_re[eb];
+        float4 r1 = *(const float4*)&s_re[eb + 4];
+        float4 i0 = *(const float4*)&s_im[eb];
+        float4 i1 = *(const float4*)&s_im[eb + 4];
+        constexpr int NG = (TP - 1) / 4; /* 4 taps per group */
+#pragma unroll 4
+        for (int m = 0; m < NG; m++) {
+            const float4 rn = *(const float4*)&s_re[eb + 4 * m + 8];
+            const float4 in_ = *(const float4*)&s_im[eb + 4 * m + 8];
+            const float4 h4 = *(const float4*)&rtaps[4 * m];
+            const float wr[8] = {r0.x, r0.y, r0.z, r0.w,
+                                 r1.x, r1.y, r1.z, r1.w};
+            const float wi[8] = {i0.x, i0.y, i0.z, i0.w,
+                                 i1.x, i1.y, i1.z, i1.w};
+            const float ht[4] = {h4.x, h4.y, h4.z, h4.w};
+#pragma unroll
+            for (int tl = 0; tl < 4; tl++) {
+#pragma unroll
+                for (int j = 0; j < FIR_R; j++) {
+                    ar[j] = fmaf(wr[j + tl], ht[tl], ar[j]);
+                    ai[j] = fmaf(wi[j + tl], ht[tl], ai[j]);
+                }
+            }
+            r0 = r1; r1 = rn;
+            i0 = i1; i1 = in_;
+        }
+        { /* final tap TP-1: r0 now holds elements TP-1..TP+2 */
+            const float h = rtaps[TP - 1];
+            const float wr[4] = {r0.x, r0.y, r0.z, r0.w};
+            const float wi[4] = {i0.x, i0.y, i0.z, i0.w};
+#pragma unroll
+            for (int j = 0; j < FIR_R; j++) {
+                ar[j] = fmaf(wr[j], h, ar[j]);
+                ai[j] = fmaf(wi[j], h, ai[j]);
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < FIR_R; j++) {
+            long long o = out_base + eb + j;
+            if (o < n_out) out[o] = make_float2(ar[j], ai[j]);
+        }
+        __syncthreads();
+    }
+}
+
+/* instantiation dispatch for the templated FIR (host side below) */
+typedef void (*fir_tpl_fn)(const float2*, float2*, const float*, long long,
+                           long long);
+template <int TP>
+static fir_tpl_fn fir_tpl_ptr() { return k_fir_cf32_tpl<TP>; }
+
 /* ================= Decimating FIR (D=4 fast path), cf32 x f32 ========= *
  * Restates decimating_fir.rs:80-95 for Complex<f32>/f32: y[k] =
  * sum_t x[D-1 + k*D + t] * h[T-1-t].
@@ -595,6 +681,8 @@ struct fsdr_filter {
     float norm = 0.f;
     int n_taps_padded = 0;   /* device taps length (leading zeros) */
     float* d_taps = nullptr;
+    int tp_tpl = 0;          /* template tap count (reversed taps) or 0 */
+    float* d_rtaps = nullptr; /* reversed taps zero-filled to tp_tpl */
     float2* d_twid = nullptr;
     /* staging buffers for the host-span path */
     void* d_in = nullptr;
@@ -647,6 +735,23 @@ extern "C" fsdr_filter* fsdr_fir_cf32_create(const float* taps,
     if (upload_taps_padded(f, taps, n_taps, 8) != FSDR_OK) {
         delete f;
         return nullptr;
+    }
+    /* template variant: reversed taps zero-filled to the smallest
+     * instantiated TP >= n_taps */
+    static const int tps[] = {17, 33, 65, 129, 257, 513};
+    for (int t : tps) {
+        if ((size_t)t >= n_taps) { f->tp_tpl = t; break; }
+    }
+    if (f->tp_tpl) {
+        std::vector<float> rt(f->tp_tpl, 0.f);
+        for (size_t i = 0; i < n_taps; i++) rt[i] = taps[n_taps - 1 - i];
+        if (hipMalloc(&f->d_rtaps, rt.size() * sizeof(float)) != hipSuccess ||
+            hipMemcpy(f->d_rtaps, rt.data(), rt.size() * sizeof(float),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_err("reversed taps upload failed");
+            delete f;
+            return nullptr;
+        }
     }
     return f;
 }
@@ -758,6 +863,7 @@ extern "C" size_t fsdr_filter_length(const fsdr_filter* f) {
 extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (!f) return;
     if (f->d_taps) (void)hipFree(f->d_taps);
+    if (f->d_rtaps) (void)hipFree(f->d_rtaps);
     if (f->d_twid) (void)hipFree(f->d_twid);
     if (f->d_in) (void)hipFree(f->d_in);
     if (f->d_out) (void)hipFree(f->d_out);
@@ -777,11 +883,30 @@ static int grid_for(long long work_items, int block) {
 static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
                            size_t n_out, size_t n_in, hipStream_t st) {
     if (n_out == 0) return FSDR_OK;
+    long long tiles = ((long long)n_out + FIR_TILE_OUT - 1) / FIR_TILE_OUT;
+    int grid = (int)std::min<long long>(tiles, 256 * 16);
+    if (f->tp_tpl) {
+        unsigned elems = FIR_TILE_OUT + f->tp_tpl + 8;
+        size_t lds = 2 * (size_t)((elems + 7u) & ~7u) * sizeof(float);
+#define FIR_TPL_CASE(TPV)                                                        case TPV:                                                                        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_fir_cf32_tpl<TPV>),                                        dim3(grid), dim3(FIR_BLOCK), lds, st,                                        (const float2*)d_in, (float2*)d_out, f->d_rtaps,                             (long long)n_out, (long long)n_in);                       break;
+        switch (f->tp_tpl) {
+            FIR_TPL_CASE(17)
+            FIR_TPL_CASE(33)
+            FIR_TPL_CASE(65)
+            FIR_TPL_CASE(129)
+            FIR_TPL_CASE(257)
+            FIR_TPL_CASE(513)
+            default:
+                set_err("bad template tap count");
+                return FSDR_ERR_INVALID;
+        }
+#undef FIR_TPL_CASE
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
+    }
     unsigned elems = FIR_TILE_OUT + f->n_taps_padded - 1 + 6;
     size_t lds = (2 * (size_t)plane_floats(elems) + f->n_taps_padded + 4) *
                  sizeof(float);
-    long long tiles = ((long long)n_out + FIR_TILE_OUT - 1) / FIR_TILE_OUT;
-    int grid = (int)std::min<long long>(tiles, 256 * 16);
     hipLaunchKernelGGL(k_fir_cf32, dim3(grid), dim3(FIR_BLOCK), lds, st,
                        (const float2*)d_in, (float2*)d_out, f->d_taps,
                        f->n_taps_padded, (long long)n_out, (long long)n_in);
