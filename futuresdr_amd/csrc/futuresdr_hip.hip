@@ -223,12 +223,13 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32(
  * Same math as k_fir_cf32; TP is the padded tap count (TP % 4 == 1) and
  * the device tap array is REVERSED (rt[i] = h[n_taps-1-i], zero-filled to
  * TP). Linear SoA LDS planes (no pad): the window is read as float4
- * groups at 16 B lane stride (the conflict-free ds_read_b128 pattern),
- * addresses are affine in the unrolled loop (fold to immediate offsets),
- * and full unroll renames the 2-group register window (no rotation movs
- * -- the v2 kernel spent ~2/3 of its VALU on movs + padded-address math,
- * profiles/rocprof_r01*). Taps come from global memory with constant
- * offsets (scalar-cache loads, keeps LDS bandwidth for the window). */
+ * groups at 16 B lane stride (the conflict-free ds_read_b128 pattern) and
+ * addresses are affine (fold to ds_read immediate offsets). Reversed taps
+ * are staged in LDS too (keeping SMEM out of the loop: s_load shares
+ * lgkmcnt with ds_read and forces lgkmcnt(0) drains). Accumulators are
+ * float2 OUTPUT pairs so every v_pk_fma reads two adjacent float4
+ * components — no register-pairing movs (the v3 variant spent ~40% of
+ * VALU on such movs). */
 template <int TP>
 __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
     const float2* __restrict__ in, float2* __restrict__ out,
@@ -238,6 +239,7 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* s_re = (float*)smem;
     float* s_im = s_re + ((elems + 7u) & ~7u);
+    float* s_rt = s_im + ((elems + 7u) & ~7u); /* TP+3 floats */
 
     const int tid = threadIdx.x;
     for (long long tile = blockIdx.x;
@@ -249,11 +251,13 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
             s_re[i] = v.x;
             s_im[i] = v.y;
         }
+        for (int i = tid; i < TP + 3; i += FIR_BLOCK)
+            s_rt[i] = (i < TP) ? rtaps[i] : 0.f;
         __syncthreads();
 
         const unsigned eb = (unsigned)tid * FIR_R;
-        float ar[FIR_R] = {0.f, 0.f, 0.f, 0.f};
-        float ai[FIR_R] = {0.f, 0.f, 0.f, 0.f};
+        float2 a01r = make_float2(0.f, 0.f), a23r = a01r;
+        float2 a01i = a01r, a23i = a01r;
         float4 r0 = *(const float4*)&s_re[eb];
         float4 r1 = *(const float4*)&s_re[eb + 4];
         float4 i0 = *(const float4*)&s_im[eb];
@@ -263,7 +267,7 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
         for (int m = 0; m < NG; m++) {
             const float4 rn = *(const float4*)&s_re[eb + 4 * m + 8];
             const float4 in_ = *(const float4*)&s_im[eb + 4 * m + 8];
-            const float4 h4 = *(const float4*)&rtaps[4 * m];
+            const float4 h4 = *(const float4*)&s_rt[4 * m];
             const float wr[8] = {r0.x, r0.y, r0.z, r0.w,
                                  r1.x, r1.y, r1.z, r1.w};
             const float wi[8] = {i0.x, i0.y, i0.z, i0.w,
@@ -271,25 +275,32 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
             const float ht[4] = {h4.x, h4.y, h4.z, h4.w};
 #pragma unroll
             for (int tl = 0; tl < 4; tl++) {
-#pragma unroll
-                for (int j = 0; j < FIR_R; j++) {
-                    ar[j] = fmaf(wr[j + tl], ht[tl], ar[j]);
-                    ai[j] = fmaf(wi[j + tl], ht[tl], ai[j]);
-                }
+                const float h = ht[tl];
+                a01r.x = fmaf(wr[tl], h, a01r.x);
+                a01r.y = fmaf(wr[tl + 1], h, a01r.y);
+                a23r.x = fmaf(wr[tl + 2], h, a23r.x);
+                a23r.y = fmaf(wr[tl + 3], h, a23r.y);
+                a01i.x = fmaf(wi[tl], h, a01i.x);
+                a01i.y = fmaf(wi[tl + 1], h, a01i.y);
+                a23i.x = fmaf(wi[tl + 2], h, a23i.x);
+                a23i.y = fmaf(wi[tl + 3], h, a23i.y);
             }
             r0 = r1; r1 = rn;
             i0 = i1; i1 = in_;
         }
         { /* final tap TP-1: r0 now holds elements TP-1..TP+2 */
-            const float h = rtaps[TP - 1];
-            const float wr[4] = {r0.x, r0.y, r0.z, r0.w};
-            const float wi[4] = {i0.x, i0.y, i0.z, i0.w};
-#pragma unroll
-            for (int j = 0; j < FIR_R; j++) {
-                ar[j] = fmaf(wr[j], h, ar[j]);
-                ai[j] = fmaf(wi[j], h, ai[j]);
-            }
+            const float h = s_rt[TP - 1];
+            a01r.x = fmaf(r0.x, h, a01r.x);
+            a01r.y = fmaf(r0.y, h, a01r.y);
+            a23r.x = fmaf(r0.z, h, a23r.x);
+            a23r.y = fmaf(r0.w, h, a23r.y);
+            a01i.x = fmaf(i0.x, h, a01i.x);
+            a01i.y = fmaf(i0.y, h, a01i.y);
+            a23i.x = fmaf(i0.z, h, a23i.x);
+            a23i.y = fmaf(i0.w, h, a23i.y);
         }
+        const float ar[4] = {a01r.x, a01r.y, a23r.x, a23r.y};
+        const float ai[4] = {a01i.x, a01i.y, a23i.x, a23i.y};
 #pragma unroll
         for (int j = 0; j < FIR_R; j++) {
             long long o = out_base + eb + j;
@@ -887,7 +898,8 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
     int grid = (int)std::min<long long>(tiles, 256 * 16);
     if (f->tp_tpl) {
         unsigned elems = FIR_TILE_OUT + f->tp_tpl + 8;
-        size_t lds = 2 * (size_t)((elems + 7u) & ~7u) * sizeof(float);
+        size_t lds = (2 * (size_t)((elems + 7u) & ~7u) + f->tp_tpl + 3) *
+                     sizeof(float);
 #define FIR_TPL_CASE(TPV)                                                        case TPV:                                                                        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_fir_cf32_tpl<TPV>),                                        dim3(grid), dim3(FIR_BLOCK), lds, st,                                        (const float2*)d_in, (float2*)d_out, f->d_rtaps,                             (long long)n_out, (long long)n_in);                       break;
         switch (f->tp_tpl) {
             FIR_TPL_CASE(17)
