@@ -262,17 +262,20 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
         float4 r1 = *(const float4*)&s_re[eb + 4];
         float4 i0 = *(const float4*)&s_im[eb];
         float4 i1 = *(const float4*)&s_im[eb + 4];
+        float4 hc = *(const float4*)&s_rt[0];
         constexpr int NG = (TP - 1) / 4; /* 4 taps per group */
+        /* software pipeline: every load issued one group ahead of its use,
+         * so in steady state no FMA waits on a just-issued ds_read */
 #pragma unroll 4
         for (int m = 0; m < NG; m++) {
             const float4 rn = *(const float4*)&s_re[eb + 4 * m + 8];
             const float4 in_ = *(const float4*)&s_im[eb + 4 * m + 8];
-            const float4 h4 = *(const float4*)&s_rt[4 * m];
+            const float4 h4 = *(const float4*)&s_rt[4 * m + 4];
             const float wr[8] = {r0.x, r0.y, r0.z, r0.w,
                                  r1.x, r1.y, r1.z, r1.w};
             const float wi[8] = {i0.x, i0.y, i0.z, i0.w,
                                  i1.x, i1.y, i1.z, i1.w};
-            const float ht[4] = {h4.x, h4.y, h4.z, h4.w};
+            const float ht[4] = {hc.x, hc.y, hc.z, hc.w};
 #pragma unroll
             for (int tl = 0; tl < 4; tl++) {
                 const float h = ht[tl];
@@ -287,9 +290,10 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
             }
             r0 = r1; r1 = rn;
             i0 = i1; i1 = in_;
+            hc = h4;
         }
-        { /* final tap TP-1: r0 now holds elements TP-1..TP+2 */
-            const float h = s_rt[TP - 1];
+        { /* final tap TP-1: r0 = elements TP-1..TP+2, hc = rt[TP-1..] */
+            const float h = hc.x;
             a01r.x = fmaf(r0.x, h, a01r.x);
             a01r.y = fmaf(r0.y, h, a01r.y);
             a23r.x = fmaf(r0.z, h, a23r.x);
