@@ -19,6 +19,7 @@
 #include <condition_variable>
 #include <cstdio>
 #include <cstring>
+#include <cstdlib>
 #include <deque>
 #include <mutex>
 #include <vector>
@@ -899,7 +900,12 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
                            size_t n_out, size_t n_in, hipStream_t st) {
     if (n_out == 0) return FSDR_OK;
     long long tiles = ((long long)n_out + FIR_TILE_OUT - 1) / FIR_TILE_OUT;
-    int grid = (int)std::min<long long>(tiles, 256 * 16);
+    /* one tile per block by default: independent blocks overlap their
+     * staging latency with other blocks' compute (a per-block tile loop
+     * stalls all 4 waves of the block at each staging barrier) */
+    long long cap = 256 * 16;
+    if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
+    int grid = (int)std::min<long long>(tiles, cap);
     if (f->tp_tpl) {
         unsigned elems = FIR_TILE_OUT + f->tp_tpl + 8;
         size_t lds = (2 * (size_t)((elems + 7u) & ~7u) + f->tp_tpl + 3) *
