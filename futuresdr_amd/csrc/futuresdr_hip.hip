@@ -424,6 +424,107 @@ __global__ __launch_bounds__(DFIR_BLOCK) void k_fir_decim4_cf32(
     }
 }
 
+/* ---- Phase-split decimating FIR, D=4, compile-time taps -------------- *
+ * Same math as decimating_fir.rs:80-95 (D=4): y[k] = sum_t x[3+4k+t] *
+ * h[T-1-t]. Decompose t = 4u+v: y[k] = sum_v sum_u P_v[k+u] * rt[4u+v]
+ * with P_v[i] = x[3+v+4i] and rt[t] = h[T-1-t] — four stride-1 sub-FIRs
+ * over de-interleaved phase planes, staged at write time (free). Each
+ * sub-FIR then uses the exact k_fir_cf32_tpl structure: conflict-free
+ * 16 B lane-stride ds_read_b128 groups, affine offsets, pipelined loads.
+ * TPD = per-phase padded tap count (multiple of 4, 4*TPD >= T). Plane
+ * stride is 8 mod 32 dwords so the de-interleaving writes are
+ * bank-conflict-free. */
+#define DFIRT_BLOCK 256
+#define DFIRT_R 4
+#define DFIRT_TILE (DFIRT_BLOCK * DFIRT_R) /* 1024 decimated outputs */
+
+__device__ __host__ __forceinline__ unsigned dfirt_sp(int tpd) {
+    return ((((unsigned)(DFIRT_TILE + tpd + 12)) + 31u) & ~31u) + 8u;
+}
+
+template <int TPD>
+__global__ __launch_bounds__(DFIRT_BLOCK) void k_fir_decim4_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv /* [4][TPD+4], rtv[v][u]=rt[4u+v] */,
+    long long n_out, long long n_in_valid) {
+    static_assert(TPD % 4 == 0, "TPD must be a multiple of 4");
+    const unsigned SP = dfirt_sp(TPD);
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* planes = (float*)smem; /* [8][SP]: re0..re3, im0..im3 */
+    float* s_rt = planes + 8u * SP; /* [4][TPD+4] */
+
+    const int tid = threadIdx.x;
+    const unsigned span = 3 + 4 * (DFIRT_TILE + TPD + 8); /* input elems */
+    for (long long tile = blockIdx.x;
+         tile * (long long)DFIRT_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * DFIRT_TILE;
+        const long long in_base = out_base * 4;
+        for (unsigned rel = 3 + tid; rel < span; rel += DFIRT_BLOCK) {
+            long long g = in_base + rel;
+            float2 x = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+            unsigned v = (rel - 3) & 3u, i = (rel - 3) >> 2;
+            planes[v * SP + i] = x.x;
+            planes[(4 + v) * SP + i] = x.y;
+        }
+        for (int i = tid; i < 4 * (TPD + 4); i += DFIRT_BLOCK)
+            s_rt[i] = rtv[i];
+        __syncthreads();
+
+        const unsigned eb = (unsigned)tid * DFIRT_R;
+        float2 a01r = make_float2(0.f, 0.f), a23r = a01r;
+        float2 a01i = a01r, a23i = a01r;
+#pragma unroll
+        for (int v = 0; v < 4; v++) {
+            const float* pre = planes + (unsigned)v * SP;
+            const float* pim = planes + (unsigned)(4 + v) * SP;
+            const float* rt = s_rt + (unsigned)v * (TPD + 4);
+            float4 r0 = *(const float4*)&pre[eb];
+            float4 r1 = *(const float4*)&pre[eb + 4];
+            float4 i0 = *(const float4*)&pim[eb];
+            float4 i1 = *(const float4*)&pim[eb + 4];
+            float4 hc = *(const float4*)&rt[0];
+            constexpr int NG = TPD / 4;
+#pragma unroll 4
+            for (int m = 0; m < NG; m++) {
+                const float4 rn = *(const float4*)&pre[eb + 4 * m + 8];
+                const float4 in_ = *(const float4*)&pim[eb + 4 * m + 8];
+                const float4 h4 = *(const float4*)&rt[4 * m + 4];
+                const float wr[8] = {r0.x, r0.y, r0.z, r0.w,
+                                     r1.x, r1.y, r1.z, r1.w};
+                const float wi[8] = {i0.x, i0.y, i0.z, i0.w,
+                                     i1.x, i1.y, i1.z, i1.w};
+                const float ht[4] = {hc.x, hc.y, hc.z, hc.w};
+#pragma unroll
+                for (int tl = 0; tl < 4; tl++) {
+                    const float h = ht[tl];
+                    a01r.x = fmaf(wr[tl], h, a01r.x);
+                    a01r.y = fmaf(wr[tl + 1], h, a01r.y);
+                    a23r.x = fmaf(wr[tl + 2], h, a23r.x);
+                    a23r.y = fmaf(wr[tl + 3], h, a23r.y);
+                    a01i.x = fmaf(wi[tl], h, a01i.x);
+                    a01i.y = fmaf(wi[tl + 1], h, a01i.y);
+                    a23i.x = fmaf(wi[tl + 2], h, a23i.x);
+                    a23i.y = fmaf(wi[tl + 3], h, a23i.y);
+                }
+                r0 = r1; r1 = rn;
+                i0 = i1; i1 = in_;
+                hc = h4;
+            }
+        }
+        const float ar[4] = {a01r.x, a01r.y, a23r.x, a23r.y};
+        const float ai[4] = {a01i.x, a01i.y, a23i.x, a23i.y};
+#pragma unroll
+        for (int j = 0; j < DFIRT_R; j++) {
+            long long o = out_base + eb + j;
+            if (o < n_out) out[o] = make_float2(ar[j], ai[j]);
+        }
+        __syncthreads();
+    }
+}
+
+typedef void (*dfir_tpl_fn)(const float2*, float2*, const float*, long long,
+                            long long);
+
 /* Generic decimating FIR (any D) — correctness fallback: one output per
  * lane per iteration, direct reads through L1/L2 (no LDS staging). */
 __global__ void k_fir_decim_generic_cf32(const float2* __restrict__ in,
@@ -802,10 +903,33 @@ extern "C" fsdr_filter* fsdr_decim_fir_cf32_create(size_t decimation,
     if (!f) return nullptr;
     f->n_taps = n_taps;
     f->decim = decimation;
-    int mod = (decimation == 4) ? 16 : 1; /* fast path needs tp%14==1 */
+    int mod = (decimation == 4) ? 16 : 1; /* fast path needs tp%16==1 */
     if (upload_taps_padded(f, taps, n_taps, mod) != FSDR_OK) {
         delete f;
         return nullptr;
+    }
+    if (decimation == 4 && n_taps <= 512) {
+        /* phase-split template: rtv[v][u] = h[n_taps-1-(4u+v)] */
+        static const int tpds[] = {8, 16, 32, 64, 128};
+        for (int t : tpds) {
+            if ((size_t)(4 * t) >= n_taps) { f->tp_tpl = t; break; }
+        }
+        int tpd = f->tp_tpl;
+        std::vector<float> rtv(4 * (tpd + 4), 0.f);
+        for (int v = 0; v < 4; v++)
+            for (int u = 0; u < tpd; u++) {
+                size_t t = 4 * (size_t)u + v;
+                if (t < n_taps)
+                    rtv[v * (tpd + 4) + u] = taps[n_taps - 1 - t];
+            }
+        if (hipMalloc(&f->d_rtaps, rtv.size() * sizeof(float)) !=
+                hipSuccess ||
+            hipMemcpy(f->d_rtaps, rtv.data(), rtv.size() * sizeof(float),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_err("rtv upload failed");
+            delete f;
+            return nullptr;
+        }
     }
     return f;
 }
@@ -939,6 +1063,28 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
 static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
                              size_t n_out, size_t n_in, hipStream_t st) {
     if (n_out == 0) return FSDR_OK;
+    if (f->decim == 4 && f->tp_tpl) {
+        long long tiles = ((long long)n_out + DFIRT_TILE - 1) / DFIRT_TILE;
+        long long cap = 256 * 64;
+        if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
+        int grid = (int)std::min<long long>(tiles, cap);
+        size_t lds = (8 * (size_t)dfirt_sp(f->tp_tpl) +
+                      4 * ((size_t)f->tp_tpl + 4)) * sizeof(float);
+#define DFIR_TPL_CASE(TPV)                                                       case TPV:                                                                        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_fir_decim4_tpl<TPV>),                                      dim3(grid), dim3(DFIRT_BLOCK), lds, st,                                      (const float2*)d_in, (float2*)d_out, f->d_rtaps,                             (long long)n_out, (long long)n_in);                       break;
+        switch (f->tp_tpl) {
+            DFIR_TPL_CASE(8)
+            DFIR_TPL_CASE(16)
+            DFIR_TPL_CASE(32)
+            DFIR_TPL_CASE(64)
+            DFIR_TPL_CASE(128)
+            default:
+                set_err("bad decim template tap count");
+                return FSDR_ERR_INVALID;
+        }
+#undef DFIR_TPL_CASE
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
+    }
     if (f->decim == 4) {
         unsigned elems = DFIR_TILE_OUT * 4 + f->n_taps_padded - 1 + 20;
         size_t lds = (2 * (size_t)plane_floats(elems) + f->n_taps_padded + 4)
