@@ -613,14 +613,24 @@ __global__ void k_resamp_cf32(const float2* __restrict__ in,
     }
 }
 
-/* ================= FFT: radix-2 Stockham in LDS ======================= *
+/* ================= FFT: radix-4 Stockham in LDS ======================= *
  * Unnormalized DFT, rustfft convention (forward e^{-2pi i kn/N}) — the
- * reference Fft block's math (src/blocks/fft.rs:190-194). One or more
- * frames per 256-thread block, ping-pong LDS, twiddle table (float2,
- * W[k] = e^{-2pi i k / N}, k < N/2) precomputed on host in f64.
- * Supports fft_shift and normalize per fft.rs:179-210. */
+ * reference Fft block's math (src/blocks/fft.rs:190-194). Stockham
+ * autosort, radix-4 stages (plus one radix-2 stage when log2 N is odd):
+ * half the LDS round trips and barriers of radix-2. One or more frames
+ * per 256-thread block, ping-pong LDS, twiddle table (float2,
+ * W[k] = e^{-2pi i k/N}, k < N, f64-computed on host). Supports
+ * fft_shift / normalize per fft.rs:179-210, and an optional fused |X|^2
+ * output (the spectrum Apply stage, examples/spectrum cpu.rs:21-28).
+ */
+__device__ __forceinline__ float2 cmul_tw(float2 a, float2 w, int inverse) {
+    if (inverse) w.y = -w.y;
+    return cmulf(a, w);
+}
+
 __global__ __launch_bounds__(256) void k_fft_stockham(
     const float2* __restrict__ in, float2* __restrict__ out,
+    float* __restrict__ mag_out /* nullable */,
     const float2* __restrict__ twid, int n, int log2n, int frames_per_block,
     int inverse, int fft_shift, float norm /* 0 = none */,
     long long n_frames) {
@@ -648,23 +658,55 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
         }
         __syncthreads();
         int scur = 1;
-        for (int stage = 0; stage < log2n; stage++) {
-            const int ncur = n >> stage;
-            const int m = ncur >> 1;
+        int ncur = n;
+        while (ncur >= 4) {
+            const int m4 = ncur >> 2;
+            if (frame < n_frames) {
+                for (int bf = tf; bf < n / 4; bf += tpf) {
+                    const int p = bf / scur;
+                    const int q = bf - p * scur;
+                    const int tw = n / ncur; /* twiddle stride */
+                    float2 x0 = a[q + scur * p];
+                    float2 x1 = a[q + scur * (p + m4)];
+                    float2 x2 = a[q + scur * (p + 2 * m4)];
+                    float2 x3 = a[q + scur * (p + 3 * m4)];
+                    /* t_k = x_k * W_ncur^(k*p) */
+                    float2 t1 = cmul_tw(x1, twid[(size_t)p * tw], inverse);
+                    float2 t2 = cmul_tw(x2, twid[(size_t)2 * p * tw],
+                                        inverse);
+                    float2 t3 = cmul_tw(x3, twid[(size_t)3 * p * tw],
+                                        inverse);
+                    /* 4-point DFT of (x0, t1, t2, t3); omega4 = -i (fwd) */
+                    float2 e0 = f2_add(x0, t2), e1 = f2_sub(x0, t2);
+                    float2 o0 = f2_add(t1, t3), o1 = f2_sub(t1, t3);
+                    /* o1 * (-i) = (o1.y, -o1.x); * (+i) for inverse */
+                    float2 o1r = inverse ? make_float2(-o1.y, o1.x)
+                                         : make_float2(o1.y, -o1.x);
+                    b[q + scur * (4 * p + 0)] = f2_add(e0, o0);
+                    b[q + scur * (4 * p + 1)] = f2_add(e1, o1r);
+                    b[q + scur * (4 * p + 2)] = f2_sub(e0, o0);
+                    b[q + scur * (4 * p + 3)] = f2_sub(e1, o1r);
+                }
+            }
+            float2* t = a; a = b; b = t;
+            scur <<= 2;
+            ncur >>= 2;
+            __syncthreads();
+        }
+        if (ncur == 2) { /* final radix-2 stage when log2n is odd */
             if (frame < n_frames) {
                 for (int bf = tf; bf < n / 2; bf += tpf) {
                     const int p = bf / scur;
                     const int q = bf - p * scur;
                     float2 xa = a[q + scur * p];
-                    float2 xb = a[q + scur * (p + m)];
-                    float2 w = twid[(size_t)p * (n / ncur)];
-                    if (inverse) w.y = -w.y;
-                    b[q + scur * 2 * p] = f2_add(xa, xb);
-                    b[q + scur * (2 * p + 1)] = cmulf(f2_sub(xa, xb), w);
+                    float2 xb = a[q + scur * (p + 1)];
+                    float2 t = cmul_tw(xb, twid[(size_t)p * (n / 2)],
+                                       inverse);
+                    b[q + scur * 2 * p] = f2_add(xa, t);
+                    b[q + scur * (2 * p + 1)] = f2_sub(xa, t);
                 }
             }
             float2* t = a; a = b; b = t;
-            scur <<= 1;
             __syncthreads();
         }
         if (frame < n_frames) {
@@ -674,6 +716,7 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                 float2 v = a[shift_out ? (i + n / 2) % n : i];
                 if (norm != 0.f) { v.x *= norm; v.y *= norm; }
                 dst[i] = v;
+                if (mag_out) mag_out[frame * n + i] = v.x * v.x + v.y * v.y;
             }
         }
         __syncthreads();
@@ -973,9 +1016,10 @@ extern "C" fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse,
     f->fft_shift = fft_shift;
     f->norm = normalize ? *normalize : 0.f;
     f->n_taps = len; /* length() = min_items = len (fft.rs:106-109) */
-    /* twiddle table W[k] = e^{-2πik/len}, k < len/2, computed in f64 */
-    std::vector<float2> tw(len / 2);
-    for (size_t k = 0; k < len / 2; k++) {
+    /* twiddle table W[k] = e^{-2πik/len}, k < len, computed in f64
+     * (radix-4 needs indices up to 3(len/4-1)) */
+    std::vector<float2> tw(len);
+    for (size_t k = 0; k < len; k++) {
         double a = -2.0 * M_PI * (double)k / (double)len;
         tw[k] = make_float2((float)cos(a), (float)sin(a));
     }
@@ -1110,7 +1154,8 @@ static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
 }
 
 static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
-                      size_t frames, hipStream_t st) {
+                      size_t frames, hipStream_t st,
+                      float* d_mag = nullptr) {
     if (frames == 0) return FSDR_OK;
     int n = (int)f->fft_len;
     int log2n = 0;
@@ -1122,9 +1167,9 @@ static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
     long long blocks = ((long long)frames + fpb - 1) / fpb;
     int grid = (int)std::min<long long>(blocks, 256 * 16);
     hipLaunchKernelGGL(k_fft_stockham, dim3(grid), dim3(256), lds, st,
-                       (const float2*)d_in, (float2*)d_out, f->d_twid, n,
-                       log2n, fpb, f->inverse, f->fft_shift, f->norm,
-                       (long long)frames);
+                       (const float2*)d_in, (float2*)d_out, d_mag,
+                       f->d_twid, n, log2n, fpb, f->inverse, f->fft_shift,
+                       f->norm, (long long)frames);
     HIP_TRY(hipGetLastError());
     return FSDR_OK;
 }
@@ -1437,15 +1482,8 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
     if (rc) return rc;
     rc = launch_decim_cf32(c->fir2, c->d_y1, c->d_y2, prod, y1_need, st);
     if (rc) return rc;
-    rc = launch_fft(c->fft, c->d_y2, out2, frames, st);
-    if (rc) return rc;
-    if (d_mag) {
-        hipLaunchKernelGGL(k_mag2, dim3(grid_for((long long)prod, 256)),
-                           dim3(256), 0, st, out2, (float*)d_mag,
-                           (long long)prod);
-        HIP_TRY(hipGetLastError());
-    }
-    return FSDR_OK;
+    rc = launch_fft(c->fft, c->d_y2, out2, frames, st, (float*)d_mag);
+    return rc;
 }
 
 /* ================= ring (Slab-style) ================================== */
