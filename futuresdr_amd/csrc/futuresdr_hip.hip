@@ -670,22 +670,23 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                     float2 x1 = a[q + scur * (p + m4)];
                     float2 x2 = a[q + scur * (p + 2 * m4)];
                     float2 x3 = a[q + scur * (p + 3 * m4)];
-                    /* t_k = x_k * W_ncur^(k*p) */
-                    float2 t1 = cmul_tw(x1, twid[(size_t)p * tw], inverse);
-                    float2 t2 = cmul_tw(x2, twid[(size_t)2 * p * tw],
-                                        inverse);
-                    float2 t3 = cmul_tw(x3, twid[(size_t)3 * p * tw],
-                                        inverse);
-                    /* 4-point DFT of (x0, t1, t2, t3); omega4 = -i (fwd) */
-                    float2 e0 = f2_add(x0, t2), e1 = f2_sub(x0, t2);
-                    float2 o0 = f2_add(t1, t3), o1 = f2_sub(t1, t3);
-                    /* o1 * (-i) = (o1.y, -o1.x); * (+i) for inverse */
+                    /* DIF radix-4: butterflies first, output twiddles
+                     * W^p, W^2p, W^3p on frequencies 1..3 (omega4 = -i
+                     * forward, +i inverse) */
+                    float2 e0 = f2_add(x0, x2), e1 = f2_sub(x0, x2);
+                    float2 o0 = f2_add(x1, x3), o1 = f2_sub(x1, x3);
                     float2 o1r = inverse ? make_float2(-o1.y, o1.x)
                                          : make_float2(o1.y, -o1.x);
                     b[q + scur * (4 * p + 0)] = f2_add(e0, o0);
-                    b[q + scur * (4 * p + 1)] = f2_add(e1, o1r);
-                    b[q + scur * (4 * p + 2)] = f2_sub(e0, o0);
-                    b[q + scur * (4 * p + 3)] = f2_sub(e1, o1r);
+                    b[q + scur * (4 * p + 1)] =
+                        cmul_tw(f2_add(e1, o1r), twid[(size_t)p * tw],
+                                inverse);
+                    b[q + scur * (4 * p + 2)] =
+                        cmul_tw(f2_sub(e0, o0), twid[(size_t)2 * p * tw],
+                                inverse);
+                    b[q + scur * (4 * p + 3)] =
+                        cmul_tw(f2_sub(e1, o1r), twid[(size_t)3 * p * tw],
+                                inverse);
                 }
             }
             float2* t = a; a = b; b = t;
@@ -700,10 +701,10 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                     const int q = bf - p * scur;
                     float2 xa = a[q + scur * p];
                     float2 xb = a[q + scur * (p + 1)];
-                    float2 t = cmul_tw(xb, twid[(size_t)p * (n / 2)],
-                                       inverse);
-                    b[q + scur * 2 * p] = f2_add(xa, t);
-                    b[q + scur * (2 * p + 1)] = f2_sub(xa, t);
+                    b[q + scur * 2 * p] = f2_add(xa, xb);
+                    b[q + scur * (2 * p + 1)] =
+                        cmul_tw(f2_sub(xa, xb), twid[(size_t)p * (n / 2)],
+                                inverse);
                 }
             }
             float2* t = a; a = b; b = t;
