@@ -120,6 +120,30 @@ def _load():
                                        ctypes.POINTER(sz)]
     lib.fsdr_chain_destroy.argtypes = [vp]
     lib.fsdr_synchronize.restype = ctypes.c_int
+    lib.fsdr_fg_create.restype = vp
+    lib.fsdr_fg_add_null_source_cf32.restype = ctypes.c_int
+    lib.fsdr_fg_add_null_source_cf32.argtypes = [vp]
+    lib.fsdr_fg_add_vector_source_cf32.restype = ctypes.c_int
+    lib.fsdr_fg_add_vector_source_cf32.argtypes = [vp, vp, sz]
+    lib.fsdr_fg_add_head.restype = ctypes.c_int
+    lib.fsdr_fg_add_head.argtypes = [vp, ctypes.c_uint64]
+    lib.fsdr_fg_add_filter.restype = ctypes.c_int
+    lib.fsdr_fg_add_filter.argtypes = [vp, vp]
+    lib.fsdr_fg_add_null_sink.restype = ctypes.c_int
+    lib.fsdr_fg_add_null_sink.argtypes = [vp]
+    lib.fsdr_fg_add_vector_sink.restype = ctypes.c_int
+    lib.fsdr_fg_add_vector_sink.argtypes = [vp]
+    lib.fsdr_fg_stream.restype = ctypes.c_int
+    lib.fsdr_fg_stream.argtypes = [vp, ctypes.c_int, ctypes.c_int]
+    lib.fsdr_fg_run.restype = ctypes.c_int
+    lib.fsdr_fg_run.argtypes = [vp]
+    lib.fsdr_fg_n_received.restype = ctypes.c_uint64
+    lib.fsdr_fg_n_received.argtypes = [vp, ctypes.c_int]
+    lib.fsdr_fg_vector_sink_get.restype = sz
+    lib.fsdr_fg_vector_sink_get.argtypes = [vp, ctypes.c_int, vp, sz]
+    lib.fsdr_fg_destroy.argtypes = [vp]
+    lib.fsdr_filter_item_sizes.restype = sz
+    lib.fsdr_filter_item_sizes.argtypes = [vp, ctypes.POINTER(sz)]
     lib.fsdr_ring_create.restype = vp
     lib.fsdr_ring_create.argtypes = [sz, sz, sz, sz]
     lib.fsdr_ring_writer_acquire.restype = ctypes.c_int
@@ -301,6 +325,65 @@ class Chain:
             ctypes.c_void_p(stream or 0), ctypes.byref(cons),
             ctypes.byref(prod)))
         return cons.value, prod.value
+
+
+class Flowgraph:
+    """Native-driver flowgraph: mirrors Flowgraph::add + connect! + run
+    (flowgraph.rs, runtime.rs) for 1-in/1-out chains on the GPU."""
+
+    def __init__(self):
+        self._h = _load().fsdr_fg_create()
+        if not self._h:
+            raise FsdrError("flowgraph create failed (no HIP device?)")
+        self._keep = []  # keep filter objects alive
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _load().fsdr_fg_destroy(self._h)
+            self._h = None
+
+    def null_source(self):
+        return _load().fsdr_fg_add_null_source_cf32(self._h)
+
+    def vector_source(self, data):
+        data = np.ascontiguousarray(data, CF32)
+        self._keep.append(data)
+        return _load().fsdr_fg_add_vector_source_cf32(
+            self._h, ctypes.c_void_p(data.ctypes.data), data.size)
+
+    def head(self, n):
+        return _load().fsdr_fg_add_head(self._h, n)
+
+    def filter(self, f):
+        self._keep.append(f)
+        return _load().fsdr_fg_add_filter(self._h, f._h)
+
+    def null_sink(self):
+        return _load().fsdr_fg_add_null_sink(self._h)
+
+    def vector_sink(self):
+        return _load().fsdr_fg_add_vector_sink(self._h)
+
+    def stream(self, src, dst):
+        _check(_load().fsdr_fg_stream(self._h, src, dst))
+
+    def connect(self, *blocks):
+        for a, b in zip(blocks, blocks[1:]):
+            self.stream(a, b)
+
+    def run(self):
+        _check(_load().fsdr_fg_run(self._h))
+
+    def n_received(self, block):
+        return _load().fsdr_fg_n_received(self._h, block)
+
+    def sink_data(self, block, dtype=CF32):
+        lib = _load()
+        nbytes = lib.fsdr_fg_vector_sink_get(self._h, block, None, 0)
+        out = np.zeros(nbytes // np.dtype(dtype).itemsize, dtype)
+        lib.fsdr_fg_vector_sink_get(self._h, block,
+                                    ctypes.c_void_p(out.ctypes.data), nbytes)
+        return out
 
 
 def kaiser_lowpass(cutoff, transition_bw, max_ripple):

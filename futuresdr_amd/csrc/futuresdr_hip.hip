@@ -1045,6 +1045,14 @@ extern "C" size_t fsdr_filter_length(const fsdr_filter* f) {
     return f ? (f->kind == K_MAG2 ? 1 : f->n_taps) : 0;
 }
 
+/* item sizes for the flowgraph driver (fsdr_fg.cpp) */
+extern "C" size_t fsdr_filter_item_sizes(const fsdr_filter* f,
+                                         size_t* out_bytes) {
+    if (!f) { if (out_bytes) *out_bytes = 8; return 8; }
+    if (out_bytes) *out_bytes = f->item_out;
+    return f->item_in;
+}
+
 extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (!f) return;
     if (f->d_taps) (void)hipFree(f->d_taps);

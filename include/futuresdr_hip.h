@@ -185,6 +185,32 @@ int  fsdr_ring_reader_acquire(fsdr_ring* r, void** dev_ptr, size_t* items);
 int  fsdr_ring_reader_release(fsdr_ring* r);
 void fsdr_ring_destroy(fsdr_ring* r);
 
+/* ---- Flowgraph driver (native C++ harness) ---------------------------- *
+ * Minimal mirror of Flowgraph::add / Flowgraph::stream / Runtime::run for
+ * 1-in/1-out chains (src/runtime/flowgraph.rs:227-241,364-423,
+ * runtime.rs:169-215, wrapped_kernel.rs:106-229) with
+ * NullSource/VectorSource/Head/NullSink/VectorSink endpoints and
+ * GPU-filter blocks. Stream data stays resident in HBM between blocks.
+ * add_* return a block id (>= 0); connect defaults to the single
+ * output->input ports, like connect!'s defaults
+ * (crates/macros/src/lib.rs:56-60). */
+typedef struct fsdr_fg fsdr_fg;
+fsdr_fg* fsdr_fg_create(void);
+int  fsdr_fg_add_null_source_cf32(fsdr_fg* fg);
+int  fsdr_fg_add_vector_source_cf32(fsdr_fg* fg, const fsdr_cf32* data,
+                                    size_t n);
+int  fsdr_fg_add_head(fsdr_fg* fg, unsigned long long n);
+int  fsdr_fg_add_filter(fsdr_fg* fg, fsdr_filter* f);
+int  fsdr_fg_add_null_sink(fsdr_fg* fg);
+int  fsdr_fg_add_vector_sink(fsdr_fg* fg);
+int  fsdr_fg_stream(fsdr_fg* fg, int src_block, int dst_block);
+int  fsdr_fg_run(fsdr_fg* fg);
+unsigned long long fsdr_fg_n_received(fsdr_fg* fg, int block);
+size_t fsdr_fg_vector_sink_get(fsdr_fg* fg, int block, void* out,
+                               size_t cap_bytes);
+void fsdr_fg_destroy(fsdr_fg* fg);
+size_t fsdr_filter_item_sizes(const fsdr_filter* f, size_t* out_bytes);
+
 #ifdef __cplusplus
 }
 #endif

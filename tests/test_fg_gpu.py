@@ -1,0 +1,97 @@
+"""Flowgraph-driver tests (GPU): the native C++ harness reproduces the
+reference runtime behavior for chains, including perf/fir's correctness
+guard `n_received == samples - stages*(taps-1)` (perf/fir/fir.rs:97).
+"""
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+def rng(s=0):
+    return np.random.default_rng(s)
+
+
+def cplx(r, n):
+    return (r.uniform(-1, 1, (n, 2)) @ [1, 1j]).astype(np.complex64)
+
+
+def test_fg_perf_fir_shape(gpu):
+    """NullSource -> Head(samples) -> Fir(64 taps) x stages -> NullSink:
+    the perf/fir flowgraph (perf/fir/fir.rs:50-74) with its correctness
+    guard n_received == samples - stages*63 (:97)."""
+    samples, stages = 1_500_000, 3
+    taps = rng(1).uniform(-1, 1, 64).astype(np.float32)
+    fg = gpu.Flowgraph()
+    src = fg.null_source()
+    head = fg.head(samples)
+    fg.stream(src, head)
+    last = head
+    for _ in range(stages):
+        f = fg.filter(gpu.Fir(taps))
+        fg.stream(last, f)
+        last = f
+    snk = fg.null_sink()
+    fg.stream(last, snk)
+    fg.run()
+    assert fg.n_received(snk) == samples - stages * 63
+
+
+def test_fg_vector_roundtrip_vs_oracle(gpu, oracle_lib):
+    r = rng(2)
+    x = cplx(r, 300000)
+    taps = r.uniform(-1, 1, 127).astype(np.float32)
+    fg = gpu.Flowgraph()
+    src = fg.vector_source(x)
+    f = fg.filter(gpu.Fir(taps))
+    snk = fg.vector_sink()
+    fg.connect(src, f, snk)
+    fg.run()
+    got = fg.sink_data(snk)
+    ref, c, p, s = oracle_lib.fir_cf32(taps, x, x.size)
+    assert got.size == p
+    err = np.abs(got - ref).max() / max(1.0, np.abs(ref).max())
+    assert err < 1e-5
+
+
+def test_fg_full_chain_vs_oracle(gpu, oracle_lib):
+    """VectorSource -> Fir -> DecimFir(4) -> Fft(256) -> Mag2 -> VectorSink
+    (the spectrum chain, examples/spectrum/src/bin/cpu.rs:21-28 shape)."""
+    r = rng(3)
+    x = cplx(r, 4 * 256 * 6 + 600)
+    t1 = r.uniform(-1, 1, 127).astype(np.float32)
+    t2 = r.uniform(-1, 1, 127).astype(np.float32)
+    fg = gpu.Flowgraph()
+    src = fg.vector_source(x)
+    f1 = fg.filter(gpu.Fir(t1))
+    f2 = fg.filter(gpu.DecimFir(4, t2))
+    f3 = fg.filter(gpu.Fft(256))
+    f4 = fg.filter(gpu.Mag2())
+    snk = fg.vector_sink()
+    fg.connect(src, f1, f2, f3, f4, snk)
+    fg.run()
+    got = fg.sink_data(snk, np.float32)
+    spectra, _ = oracle_lib.chain_cf32(t1, t2, 4, 256, x)
+    ref = np.abs(spectra) ** 2
+    assert got.size == ref.size
+    err = np.abs(got - ref).max() / max(1.0, ref.max())
+    assert err < 1e-4
+
+
+def test_fg_small_buffer_many_chunks(gpu, oracle_lib):
+    """Stream much more data than one edge buffer holds — exercises the
+    compaction (slab tail-copy) path repeatedly."""
+    r = rng(4)
+    x = cplx(r, 1 << 20)  # 4x the default edge capacity
+    taps = r.uniform(-1, 1, 63).astype(np.float32)
+    fg = gpu.Flowgraph()
+    src = fg.vector_source(x)
+    f = fg.filter(gpu.Fir(taps))
+    snk = fg.vector_sink()
+    fg.connect(src, f, snk)
+    fg.run()
+    got = fg.sink_data(snk)
+    assert got.size == x.size - 62
+    ref, _, p, _ = oracle_lib.fir_cf32(taps, x, x.size)
+    err = np.abs(got - ref).max() / max(1.0, np.abs(ref).max())
+    assert err < 1e-5
