@@ -267,7 +267,7 @@ __global__ __launch_bounds__(FIR_BLOCK) void k_fir_cf32_tpl(
         constexpr int NG = (TP - 1) / 4; /* 4 taps per group */
         /* software pipeline: every load issued one group ahead of its use,
          * so in steady state no FMA waits on a just-issued ds_read */
-#pragma unroll 4
+#pragma unroll 8
         for (int m = 0; m < NG; m++) {
             const float4 rn = *(const float4*)&s_re[eb + 4 * m + 8];
             const float4 in_ = *(const float4*)&s_im[eb + 4 * m + 8];
@@ -450,8 +450,12 @@ __global__ __launch_bounds__(DFIRT_BLOCK) void k_fir_decim4_tpl(
     static_assert(TPD % 4 == 0, "TPD must be a multiple of 4");
     const unsigned SP = dfirt_sp(TPD);
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    float* planes = (float*)smem; /* [8][SP]: re0..re3, im0..im3 */
-    float* s_rt = planes + 8u * SP; /* [4][TPD+4] */
+    /* two phases resident at a time: [re_v0, re_v1, im_v0, im_v1] — halves
+     * the LDS footprint (8 -> 4 planes), doubling blocks/CU vs the
+     * all-phases layout (the all-phase version was LDS-capacity-bound at
+     * 4 blocks/CU). Accumulators carry across the two halves. */
+    float* planes = (float*)smem;   /* [4][SP] */
+    float* s_rt = planes + 4u * SP; /* [4][TPD+4] */
 
     const int tid = threadIdx.x;
     const unsigned span = 3 + 4 * (DFIRT_TILE + TPD + 8); /* input elems */
@@ -459,57 +463,68 @@ __global__ __launch_bounds__(DFIRT_BLOCK) void k_fir_decim4_tpl(
          tile * (long long)DFIRT_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * DFIRT_TILE;
         const long long in_base = out_base * 4;
-        for (unsigned rel = 3 + tid; rel < span; rel += DFIRT_BLOCK) {
-            long long g = in_base + rel;
-            float2 x = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
-            unsigned v = (rel - 3) & 3u, i = (rel - 3) >> 2;
-            planes[v * SP + i] = x.x;
-            planes[(4 + v) * SP + i] = x.y;
-        }
-        for (int i = tid; i < 4 * (TPD + 4); i += DFIRT_BLOCK)
-            s_rt[i] = rtv[i];
-        __syncthreads();
-
         const unsigned eb = (unsigned)tid * DFIRT_R;
         float2 a01r = make_float2(0.f, 0.f), a23r = a01r;
         float2 a01i = a01r, a23i = a01r;
 #pragma unroll
-        for (int v = 0; v < 4; v++) {
-            const float* pre = planes + (unsigned)v * SP;
-            const float* pim = planes + (unsigned)(4 + v) * SP;
-            const float* rt = s_rt + (unsigned)v * (TPD + 4);
-            float4 r0 = *(const float4*)&pre[eb];
-            float4 r1 = *(const float4*)&pre[eb + 4];
-            float4 i0 = *(const float4*)&pim[eb];
-            float4 i1 = *(const float4*)&pim[eb + 4];
-            float4 hc = *(const float4*)&rt[0];
-            constexpr int NG = TPD / 4;
-#pragma unroll 4
-            for (int m = 0; m < NG; m++) {
-                const float4 rn = *(const float4*)&pre[eb + 4 * m + 8];
-                const float4 in_ = *(const float4*)&pim[eb + 4 * m + 8];
-                const float4 h4 = *(const float4*)&rt[4 * m + 4];
-                const float wr[8] = {r0.x, r0.y, r0.z, r0.w,
-                                     r1.x, r1.y, r1.z, r1.w};
-                const float wi[8] = {i0.x, i0.y, i0.z, i0.w,
-                                     i1.x, i1.y, i1.z, i1.w};
-                const float ht[4] = {hc.x, hc.y, hc.z, hc.w};
-#pragma unroll
-                for (int tl = 0; tl < 4; tl++) {
-                    const float h = ht[tl];
-                    a01r.x = fmaf(wr[tl], h, a01r.x);
-                    a01r.y = fmaf(wr[tl + 1], h, a01r.y);
-                    a23r.x = fmaf(wr[tl + 2], h, a23r.x);
-                    a23r.y = fmaf(wr[tl + 3], h, a23r.y);
-                    a01i.x = fmaf(wi[tl], h, a01i.x);
-                    a01i.y = fmaf(wi[tl + 1], h, a01i.y);
-                    a23i.x = fmaf(wi[tl + 2], h, a23i.x);
-                    a23i.y = fmaf(wi[tl + 3], h, a23i.y);
-                }
-                r0 = r1; r1 = rn;
-                i0 = i1; i1 = in_;
-                hc = h4;
+        for (int half = 0; half < 2; half++) {
+            /* stage phases v = 2*half, 2*half+1: input elements with
+             * (rel-3) % 4 in {2h, 2h+1}; each element = P_v[i] at
+             * rel = 3 + 4i + v */
+            for (unsigned idx = tid; 2 * idx + 3 < span;
+                 idx += DFIRT_BLOCK) {
+                unsigned i = idx >> 1, vloc = idx & 1u;
+                unsigned rel = 3 + 4 * i + 2 * half + vloc;
+                if (rel >= span) continue;
+                long long g = in_base + rel;
+                float2 x = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+                planes[vloc * SP + i] = x.x;
+                planes[(2 + vloc) * SP + i] = x.y;
             }
+            if (half == 0)
+                for (int i = tid; i < 4 * (TPD + 4); i += DFIRT_BLOCK)
+                    s_rt[i] = rtv[i];
+            __syncthreads();
+#pragma unroll
+            for (int vloc = 0; vloc < 2; vloc++) {
+                const float* pre = planes + (unsigned)vloc * SP;
+                const float* pim = planes + (unsigned)(2 + vloc) * SP;
+                const float* rt =
+                    s_rt + (unsigned)(2 * half + vloc) * (TPD + 4);
+                float4 r0 = *(const float4*)&pre[eb];
+                float4 r1 = *(const float4*)&pre[eb + 4];
+                float4 i0 = *(const float4*)&pim[eb];
+                float4 i1 = *(const float4*)&pim[eb + 4];
+                float4 hc = *(const float4*)&rt[0];
+                constexpr int NG = TPD / 4;
+#pragma unroll 4
+                for (int m = 0; m < NG; m++) {
+                    const float4 rn = *(const float4*)&pre[eb + 4 * m + 8];
+                    const float4 in_ = *(const float4*)&pim[eb + 4 * m + 8];
+                    const float4 h4 = *(const float4*)&rt[4 * m + 4];
+                    const float wr[8] = {r0.x, r0.y, r0.z, r0.w,
+                                         r1.x, r1.y, r1.z, r1.w};
+                    const float wi[8] = {i0.x, i0.y, i0.z, i0.w,
+                                         i1.x, i1.y, i1.z, i1.w};
+                    const float ht[4] = {hc.x, hc.y, hc.z, hc.w};
+#pragma unroll
+                    for (int tl = 0; tl < 4; tl++) {
+                        const float h = ht[tl];
+                        a01r.x = fmaf(wr[tl], h, a01r.x);
+                        a01r.y = fmaf(wr[tl + 1], h, a01r.y);
+                        a23r.x = fmaf(wr[tl + 2], h, a23r.x);
+                        a23r.y = fmaf(wr[tl + 3], h, a23r.y);
+                        a01i.x = fmaf(wi[tl], h, a01i.x);
+                        a01i.y = fmaf(wi[tl + 1], h, a01i.y);
+                        a23i.x = fmaf(wi[tl + 2], h, a23i.x);
+                        a23i.y = fmaf(wi[tl + 3], h, a23i.y);
+                    }
+                    r0 = r1; r1 = rn;
+                    i0 = i1; i1 = in_;
+                    hc = h4;
+                }
+            }
+            __syncthreads(); /* before restaging / next tile */
         }
         const float ar[4] = {a01r.x, a01r.y, a23r.x, a23r.y};
         const float ai[4] = {a01i.x, a01i.y, a23i.x, a23i.y};
@@ -518,7 +533,6 @@ __global__ __launch_bounds__(DFIRT_BLOCK) void k_fir_decim4_tpl(
             long long o = out_base + eb + j;
             if (o < n_out) out[o] = make_float2(ar[j], ai[j]);
         }
-        __syncthreads();
     }
 }
 
@@ -1121,7 +1135,7 @@ static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
         long long cap = 256 * 64;
         if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
         int grid = (int)std::min<long long>(tiles, cap);
-        size_t lds = (8 * (size_t)dfirt_sp(f->tp_tpl) +
+        size_t lds = (4 * (size_t)dfirt_sp(f->tp_tpl) +
                       4 * ((size_t)f->tp_tpl + 4)) * sizeof(float);
 #define DFIR_TPL_CASE(TPV)                                                       case TPV:                                                                        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_fir_decim4_tpl<TPV>),                                      dim3(grid), dim3(DFIRT_BLOCK), lds, st,                                      (const float2*)d_in, (float2*)d_out, f->d_rtaps,                             (long long)n_out, (long long)n_in);                       break;
         switch (f->tp_tpl) {
