@@ -71,6 +71,13 @@ def _load():
     lib.fsdr_fir_cf32_create.argtypes = [f32p, sz]
     lib.fsdr_fir_f32_create.restype = vp
     lib.fsdr_fir_f32_create.argtypes = [f32p, sz]
+    lib.fsdr_fir_ccf32_create.restype = vp
+    lib.fsdr_fir_ccf32_create.argtypes = [vp, sz]
+    lib.fsdr_rotator_dev.restype = ctypes.c_int
+    lib.fsdr_rotator_dev.argtypes = [vp, vp, sz, ctypes.c_float,
+                                     ctypes.c_float, ctypes.c_float, vp,
+                                     ctypes.POINTER(ctypes.c_float),
+                                     ctypes.POINTER(ctypes.c_float)]
     lib.fsdr_decim_fir_cf32_create.restype = vp
     lib.fsdr_decim_fir_cf32_create.argtypes = [sz, f32p, sz]
     lib.fsdr_resamp_cf32_create.restype = vp
@@ -253,6 +260,43 @@ class Fir(Filter):
         self._taps_keep, p = _f32(taps)
         super().__init__(_load().fsdr_fir_cf32_create(p,
                                                       self._taps_keep.size))
+
+
+class FirCC(Filter):
+    """Complex-taps FIR — fir.rs:257-277 (WLAN correlator core)."""
+
+    def __init__(self, taps):
+        self._taps_keep = np.ascontiguousarray(taps, CF32)
+        super().__init__(_load().fsdr_fir_ccf32_create(
+            ctypes.c_void_p(self._taps_keep.ctypes.data),
+            self._taps_keep.size))
+
+
+def rotator_host(inp, phase_incr_angle, phase0=1.0 + 0.0j):
+    """Rotator over a host span (stages via dev helpers)."""
+    lib = _load()
+    inp = np.ascontiguousarray(inp, CF32)
+    n = inp.size
+    d_in = ctypes.c_void_p()
+    d_out = ctypes.c_void_p()
+    _check(lib.fsdr_dev_alloc(ctypes.byref(d_in), max(n, 1) * 8))
+    _check(lib.fsdr_dev_alloc(ctypes.byref(d_out), max(n, 1) * 8))
+    try:
+        _check(lib.fsdr_memcpy_h2d(d_in, ctypes.c_void_p(inp.ctypes.data),
+                                   n * 8))
+        fr = ctypes.c_float()
+        fi = ctypes.c_float()
+        _check(lib.fsdr_rotator_dev(d_in, d_out, n, phase_incr_angle,
+                                    phase0.real, phase0.imag, None,
+                                    ctypes.byref(fr), ctypes.byref(fi)))
+        _check(lib.fsdr_synchronize())
+        out = np.zeros(n, CF32)
+        _check(lib.fsdr_memcpy_d2h(ctypes.c_void_p(out.ctypes.data), d_out,
+                                   n * 8))
+        return out, complex(fr.value, fi.value)
+    finally:
+        lib.fsdr_dev_free(d_in)
+        lib.fsdr_dev_free(d_out)
 
 
 class FirF32(Filter):

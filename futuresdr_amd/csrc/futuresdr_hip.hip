@@ -563,6 +563,74 @@ __global__ void k_fir_decim_generic_cf32(const float2* __restrict__ in,
     }
 }
 
+/* ================= FIR cf32 x cf32 (complex taps) ===================== *
+ * FirFilter<Complex32,Complex32,Complex32> — fir.rs:257-277 (stable path):
+ * accum + sample*tap with the num_complex multiply. This is the WLAN
+ * SyncLong correlator core (examples/wlan/src/sync_long.rs:18-50, 64
+ * complex taps). Correctness-first kernel: LDS-staged tile, one output
+ * per lane x4 lane-strided (stride-1 LDS reads), taps staged in LDS.
+ * An MFMA/window-optimized variant is a later-round item.
+ */
+__global__ __launch_bounds__(256) void k_fir_ccf32(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float2* __restrict__ taps, int n_taps, long long n_out,
+    long long n_in_valid) {
+    const unsigned TILE = 1024;
+    const unsigned elems = TILE + n_taps - 1;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float2* s_x = (float2*)smem;          /* elems */
+    float2* s_h = s_x + elems;            /* reversed taps */
+    const int tid = threadIdx.x;
+    for (long long tile = blockIdx.x; tile * (long long)TILE < n_out;
+         tile += gridDim.x) {
+        const long long out_base = tile * TILE;
+        for (unsigned i = tid; i < elems; i += 256) {
+            long long g = out_base + i;
+            s_x[i] = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+        }
+        for (int i = tid; i < n_taps; i += 256)
+            s_h[i] = taps[n_taps - 1 - i]; /* reversed */
+        __syncthreads();
+        for (int j = 0; j < 4; j++) {
+            unsigned k = tid + j * 256;
+            float sre = 0.f, sim = 0.f;
+            for (int t = 0; t < n_taps; t++) {
+                float2 x = s_x[k + t];
+                float2 h = s_h[t];
+                sre = fmaf(x.x, h.x, sre);
+                sre = fmaf(-x.y, h.y, sre);
+                sim = fmaf(x.x, h.y, sim);
+                sim = fmaf(x.y, h.x, sim);
+            }
+            long long o = out_base + k;
+            if (o < n_out) out[o] = make_float2(sre, sim);
+        }
+        __syncthreads();
+    }
+}
+
+/* ================= Rotator ============================================ *
+ * Rotator::rotate — crates/futuredsp/src/rotator.rs:23-49: out[i] =
+ * in[i] * phase0 * e^{i*theta*(i+1)}. The reference iterates
+ * phase *= phase_incr per sample (accumulating f32 rounding); this kernel
+ * computes the phase in closed form per sample (sincosf), which tracks
+ * the IDEAL rotation — parity vs the oracle is tolerance-bounded by the
+ * oracle's own drift (documented in tests). */
+__global__ void k_rotator(const float2* __restrict__ in,
+                          float2* __restrict__ out, long long n,
+                          float theta, float p0r, float p0i) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        float s, c;
+        __sincosf(theta * (float)(i + 1), &s, &c);
+        float pr = c * p0r - s * p0i;
+        float pi = c * p0i + s * p0r;
+        float2 x = in[i];
+        out[i] = make_float2(x.x * pr - x.y * pi, x.x * pi + x.y * pr);
+    }
+}
+
 /* ================= FIR f32 x f32 (plumbing config) ==================== *
  * fir.rs:206-215 semantics; simple LDS-staged kernel (this path is the
  * reference's perf/fir plumbing shape, not the metric). */
@@ -845,7 +913,7 @@ static fsdr_filter_result resamp_status(size_t L, size_t M, size_t nt_total,
 /* ================= filter handles ===================================== */
 
 enum FilterKind { K_FIR_CF32, K_FIR_F32, K_DECIM_CF32, K_RESAMP_CF32,
-                  K_FFT_CF32, K_MAG2 };
+                  K_FFT_CF32, K_MAG2, K_FIR_CCF32 };
 
 struct fsdr_filter {
     FilterKind kind;
@@ -927,6 +995,23 @@ extern "C" fsdr_filter* fsdr_fir_cf32_create(const float* taps,
             delete f;
             return nullptr;
         }
+    }
+    return f;
+}
+
+extern "C" fsdr_filter* fsdr_fir_ccf32_create(const fsdr_cf32* taps,
+                                              size_t n_taps) {
+    if (!taps || n_taps == 0) { set_err("null/empty taps"); return nullptr; }
+    fsdr_filter* f = create_common(K_FIR_CCF32);
+    if (!f) return nullptr;
+    f->n_taps = n_taps;
+    f->n_taps_padded = (int)n_taps;
+    if (hipMalloc(&f->d_taps, n_taps * sizeof(fsdr_cf32)) != hipSuccess ||
+        hipMemcpy(f->d_taps, taps, n_taps * sizeof(fsdr_cf32),
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        set_err("taps upload failed");
+        delete f;
+        return nullptr;
     }
     return f;
 }
@@ -1249,6 +1334,20 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
             r->status = FSDR_BOTH_SUFFICIENT;
             return launch_fft(f, d_in, d_out, m / f->fft_len, st);
         }
+        case K_FIR_CCF32: {
+            *r = fir_status(n_in, f->n_taps, n_out);
+            if (r->produced == 0) return FSDR_OK;
+            unsigned elems = 1024 + f->n_taps - 1;
+            size_t lds = ((size_t)elems + f->n_taps) * sizeof(float2);
+            long long tiles = ((long long)r->produced + 1023) / 1024;
+            int grid = (int)std::min<long long>(tiles, 256 * 64);
+            hipLaunchKernelGGL(k_fir_ccf32, dim3(grid), dim3(256), lds, st,
+                               (const float2*)d_in, (float2*)d_out,
+                               (const float2*)f->d_taps, (int)f->n_taps,
+                               (long long)r->produced, (long long)n_in);
+            HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
         case K_MAG2: {
             size_t m = n_in < n_out ? n_in : n_out; /* apply.rs:108 */
             r->consumed = r->produced = m;
@@ -1429,6 +1528,27 @@ extern "C" int fsdr_firdes_lowpass_kaiser_n_f32(size_t n_taps, double beta,
     fsdr_kaiser_window(n_taps, beta, win.data());
     firdes_lowpass_h(cutoff, win.data(), n_taps, taps.data());
     for (size_t i = 0; i < n_taps; i++) out[i] = (float)taps[i];
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_rotator_dev(const void* d_in, void* d_out, size_t n,
+                                float phase_incr_angle, float phase0_re,
+                                float phase0_im, void* stream,
+                                float* final_re, float* final_im) {
+    REQUIRE_GPU();
+    if (n > 0) {
+        hipLaunchKernelGGL(k_rotator, dim3(grid_for((long long)n, 256)),
+                           dim3(256), 0, (hipStream_t)stream,
+                           (const float2*)d_in, (float2*)d_out, (long long)n,
+                           phase_incr_angle, phase0_re, phase0_im);
+        HIP_TRY(hipGetLastError());
+    }
+    if (final_re && final_im) { /* closed-form final phase (f64) */
+        double a = (double)phase_incr_angle * (double)n;
+        double c = cos(a), s = sin(a);
+        *final_re = (float)(c * phase0_re - s * phase0_im);
+        *final_im = (float)(c * phase0_im + s * phase0_re);
+    }
     return FSDR_OK;
 }
 

@@ -88,6 +88,9 @@ typedef struct fsdr_filter fsdr_filter;
 fsdr_filter* fsdr_fir_cf32_create(const float* taps, size_t n_taps);
 /* FirFilter<f32,f32,f32> — fir.rs:206-215. */
 fsdr_filter* fsdr_fir_f32_create(const float* taps, size_t n_taps);
+/* FirFilter<Complex32,Complex32,Complex32> — fir.rs:257-277 (complex
+ * taps; the WLAN SyncLong correlator core, sync_long.rs:18-50). */
+fsdr_filter* fsdr_fir_ccf32_create(const fsdr_cf32* taps, size_t n_taps);
 /* DecimatingFirFilter<Complex32,Complex32,f32> — decimating_fir.rs. */
 fsdr_filter* fsdr_decim_fir_cf32_create(size_t decimation,
                                         const float* taps, size_t n_taps);
@@ -120,6 +123,14 @@ int fsdr_cmul_dev(const void* d_a, size_t n_a, const void* d_b, size_t n_b,
                   void* d_out, size_t n_out, void* stream, size_t* m);
 int fsdr_cmul_host(const void* a, size_t n_a, const void* b, size_t n_b,
                    void* out, size_t n_out, size_t* m);
+
+/* Rotator (futuredsp rotator.rs:23-49): out[i] = in[i] * phase0 *
+ * e^{i*angle*(i+1)}, closed-form phase (the reference iterates — see the
+ * parity note in tests/test_gpu_parity.py). Returns the final phase. */
+int fsdr_rotator_dev(const void* d_in, void* d_out, size_t n,
+                     float phase_incr_angle, float phase0_re,
+                     float phase0_im, void* stream, float* final_re,
+                     float* final_im);
 
 /* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
 int fsdr_dev_alloc(void** d_ptr, size_t bytes);

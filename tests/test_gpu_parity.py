@@ -467,3 +467,58 @@ def test_filter_dev_on_torch_stream(gpu):
     ref, _, _, _ = __import__("oracle").fir_cf32(
         taps, x.cpu().numpy().view(np.complex64).ravel(), y.shape[0])
     assert_close(got, ref)
+
+
+# ---------------- complex-taps FIR (WLAN correlator core) --------------
+
+@pytest.mark.parametrize("n_taps,n_in", [(64, 320), (64, 10000), (3, 50),
+                                         (127, 5000)])
+def test_fir_ccf32_parity(gpu, oracle_lib, n_taps, n_in):
+    r = rng(n_taps * 7 + n_in)
+    taps = cplx(r, n_taps)
+    x = cplx(r, n_in)
+    got, c, p, s = gpu.FirCC(taps).filter(x, n_in)
+    ref, co, po, so = oracle_lib.fir_ccf32(taps, x, n_in)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref)
+
+
+def test_fir_ccf32_wlan_sync_long_shape(gpu, oracle_lib):
+    """64-tap complex correlator over SEARCH_WINDOW=320 samples — the
+    SyncLong shape (examples/wlan/src/sync_long.rs:18-50): correlate,
+    mag^2, and find the top-2 peak indices; compare against the oracle
+    pipeline end-to-end."""
+    r = rng(802)
+    # synthetic 'LTF-like' training sequence embedded at a known offset
+    ltf = cplx(r, 64)
+    sig = cplx(r, 320) * 0.05
+    off = 123
+    sig[off:off + 64] += ltf
+    sig[off + 64:off + 128] += ltf  # repeated symbol -> two peaks
+    taps = np.conj(ltf[::-1])  # matched filter
+    got, _, p, _ = gpu.FirCC(taps).filter(sig, 320)
+    ref, _, po, _ = oracle_lib.fir_ccf32(taps, sig, 320)
+    assert p == po
+    assert_close(got, ref)
+    mags = np.abs(got) ** 2
+    top2 = np.sort(np.argsort(mags)[-2:])
+    assert abs(int(top2[1]) - int(top2[0])) == 64  # peak spacing = symbol
+
+
+# ---------------- rotator ----------------------------------------------
+
+def test_rotator_vs_ideal_and_oracle(gpu, oracle_lib):
+    import cmath
+    r = rng(90)
+    n = 2048
+    x = cplx(r, n)
+    theta = 0.31
+    got, final = gpu.rotator_host(x, theta)
+    # vs ideal f64 rotation (the GPU computes closed-form phases)
+    ideal = x * np.exp(1j * theta * (np.arange(n, dtype=np.float64) + 1))
+    assert np.abs(got - ideal).max() < 1e-4
+    # vs the oracle (reference semantics: iterated multiply, which drifts
+    # from ideal by O(n*eps) — bound the comparison accordingly)
+    ref, ref_phase = oracle_lib.rotator(theta, x)
+    assert np.abs(got - ref).max() < 5e-4
+    assert abs(final - cmath.exp(1j * theta * n)) < 1e-5
