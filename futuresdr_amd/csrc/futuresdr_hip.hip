@@ -424,6 +424,93 @@ __global__ __launch_bounds__(DFIR_BLOCK) void k_fir_decim4_cf32(
     }
 }
 
+/* ---- MFMA FIR variant (f32-input matrix cores) ----------------------- *
+ * Same math as k_fir_cf32_tpl, reformulated as an implicit GEMM for the
+ * exact-f32 matrix cores (v_mfma_f32_16x16x4_f32 — same 157.3 TF peak as
+ * the f32 VALU but ~30x fewer instructions, so the VALU overhead and
+ * issue stalls of the FMA kernel disappear):
+ *   C[i][j] = sum_k X[i][k] * H[k][j],  X[i][k] = x[base + 16 i + k],
+ *   H[k][j] = rt[k - j]  (0 <= k-j < T else 0),  K = KK >= T+15, KK%4==0
+ * => C[i][j] = y[base + 16 i + j]; one 16x16 C tile (re + im, two
+ * accumulator chains = full MFMA issue rate) per wave per 256 outputs.
+ * The H fragments depend only on the taps: computed once per block into
+ * KK/4 VGPRs. A fragments are single ds_read_b32 per MFMA from linear
+ * SoA planes with an XOR bank swizzle (rows stride 16 dwords; lanes r,
+ * r+2m collide mod 32, so XOR bits 5..7 of the dword index into bits
+ * 2..4 — conflict-free, verified in the bank math of DESIGN.md). */
+typedef float v4f __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ unsigned mfma_swz(unsigned idx) {
+    return idx ^ (((idx >> 5) & 7u) << 2);
+}
+
+#define MFIR_BLOCK 256
+#define MFIR_TILE 1024 /* 4 waves x 256 outputs */
+
+template <int KK>
+__global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtaps /* reversed, zero-filled to KK */,
+    long long n_out, long long n_in_valid) {
+    static_assert(KK % 4 == 0, "KK must be a multiple of 4");
+    const unsigned elems = MFIR_TILE + KK + 8;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* s_re = (float*)smem;
+    float* s_im = s_re + ((elems + 31u) & ~31u);
+    float* s_rtx = s_im + ((elems + 31u) & ~31u); /* 15 zeros + rt[KK] + 1 */
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;   /* A row / C col */
+    const int k4 = lane >> 4;    /* k sub-slice 0..3 */
+
+    /* B fragments: lane holds H[k0 + k4][j = r16] = rtx[k0+k4-r16],
+     * identical for every tile -> compute once (s_rtx has a 15-zero
+     * prologue so the index is never negative). */
+    for (int i = tid; i < KK + 16; i += MFIR_BLOCK)
+        s_rtx[i] = (i >= 15 && i < 15 + KK) ? rtaps[i - 15] : 0.f;
+    __syncthreads();
+    float bfrag[KK / 4];
+#pragma unroll
+    for (int s = 0; s < KK / 4; s++)
+        bfrag[s] = s_rtx[15 + 4 * s + k4 - r16];
+    __syncthreads();
+
+    for (long long tile = blockIdx.x;
+         tile * (long long)MFIR_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MFIR_TILE;
+        for (unsigned i = tid; i < elems; i += MFIR_BLOCK) {
+            long long g = out_base + i;
+            float2 v = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+            s_re[mfma_swz(i)] = v.x;
+            s_im[mfma_swz(i)] = v.y;
+        }
+        __syncthreads();
+
+        const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int s = 0; s < KK / 4; s++) {
+            float a_re = s_re[mfma_swz(ab + 4 * s)];
+            float a_im = s_im[mfma_swz(ab + 4 * s)];
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
+                                                       0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
+                                                       0, 0, 0);
+        }
+        /* C layout: col = lane&15, row = (lane>>4)*4 + q (cdna4 16x16) */
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            long long o = out_base + (long long)wave * 256 + 16 * row + r16;
+            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+    }
+}
+
 /* ---- Phase-split decimating FIR, D=4, compile-time taps -------------- *
  * Same math as decimating_fir.rs:80-95 (D=4): y[k] = sum_t x[3+4k+t] *
  * h[T-1-t]. Decompose t = 4u+v: y[k] = sum_v sum_u P_v[k+u] * rt[4u+v]
@@ -926,6 +1013,8 @@ struct fsdr_filter {
     float* d_taps = nullptr;
     int tp_tpl = 0;          /* template tap count (reversed taps) or 0 */
     float* d_rtaps = nullptr; /* reversed taps zero-filled to tp_tpl */
+    int kk_mfma = 0;         /* MFMA variant K (>= n_taps+15, %4==0) */
+    float* d_mtaps = nullptr; /* reversed taps zero-filled to kk_mfma */
     float2* d_twid = nullptr;
     /* staging buffers for the host-span path */
     void* d_in = nullptr;
@@ -992,6 +1081,20 @@ extern "C" fsdr_filter* fsdr_fir_cf32_create(const float* taps,
             hipMemcpy(f->d_rtaps, rt.data(), rt.size() * sizeof(float),
                       hipMemcpyHostToDevice) != hipSuccess) {
             set_err("reversed taps upload failed");
+            delete f;
+            return nullptr;
+        }
+    }
+    static const int kks[] = {32, 64, 96, 144, 272, 528};
+    for (int k : kks)
+        if ((size_t)k >= n_taps + 15) { f->kk_mfma = k; break; }
+    if (f->kk_mfma) {
+        std::vector<float> rt(f->kk_mfma, 0.f);
+        for (size_t i = 0; i < n_taps; i++) rt[i] = taps[n_taps - 1 - i];
+        if (hipMalloc(&f->d_mtaps, rt.size() * sizeof(float)) != hipSuccess ||
+            hipMemcpy(f->d_mtaps, rt.data(), rt.size() * sizeof(float),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_err("mfma taps upload failed");
             delete f;
             return nullptr;
         }
@@ -1156,6 +1259,7 @@ extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (!f) return;
     if (f->d_taps) (void)hipFree(f->d_taps);
     if (f->d_rtaps) (void)hipFree(f->d_rtaps);
+    if (f->d_mtaps) (void)hipFree(f->d_mtaps);
     if (f->d_twid) (void)hipFree(f->d_twid);
     if (f->d_in) (void)hipFree(f->d_in);
     if (f->d_out) (void)hipFree(f->d_out);
@@ -1182,6 +1286,27 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
     long long cap = 256 * 16;
     if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
     int grid = (int)std::min<long long>(tiles, cap);
+    const char* mf = getenv("FSDR_FIR_MFMA");
+    if (f->kk_mfma && mf && atoi(mf) != 0) {
+        unsigned elems = MFIR_TILE + f->kk_mfma + 8;
+        size_t lds = (2 * (size_t)((elems + 31u) & ~31u) + f->kk_mfma + 20)
+                     * sizeof(float);
+#define MFIR_TPL_CASE(KV)                                                        case KV:                                                                         hipLaunchKernelGGL(HIP_KERNEL_NAME(k_fir_mfma_tpl<KV>), dim3(grid),                             dim3(MFIR_BLOCK), lds, st, (const float2*)d_in,                              (float2*)d_out, f->d_mtaps, (long long)n_out,                                (long long)n_in);                                         break;
+        switch (f->kk_mfma) {
+            MFIR_TPL_CASE(32)
+            MFIR_TPL_CASE(64)
+            MFIR_TPL_CASE(96)
+            MFIR_TPL_CASE(144)
+            MFIR_TPL_CASE(272)
+            MFIR_TPL_CASE(528)
+            default:
+                set_err("bad mfma K");
+                return FSDR_ERR_INVALID;
+        }
+#undef MFIR_TPL_CASE
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
+    }
     if (f->tp_tpl) {
         unsigned elems = FIR_TILE_OUT + f->tp_tpl + 8;
         size_t lds = (2 * (size_t)((elems + 7u) & ~7u) + f->tp_tpl + 3) *
