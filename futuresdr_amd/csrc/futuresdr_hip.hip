@@ -480,11 +480,17 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)MFIR_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * MFIR_TILE;
-        for (unsigned i = tid; i < elems; i += MFIR_BLOCK) {
+        /* 2 complex per thread-iteration: float4 global load, paired b64
+         * LDS writes (the XOR swizzle preserves even-pair adjacency) */
+        for (unsigned i = 2 * tid; i < elems; i += 2 * MFIR_BLOCK) {
             long long g = out_base + i;
-            float2 v = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
-            s_re[mfma_swz(i)] = v.x;
-            s_im[mfma_swz(i)] = v.y;
+            float2 v0 = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+            float2 v1 = (g + 1 < n_in_valid && i + 1 < elems)
+                            ? in[g + 1]
+                            : make_float2(0.f, 0.f);
+            unsigned d = mfma_swz(i);
+            *(float2*)&s_re[d] = make_float2(v0.x, v1.x);
+            *(float2*)&s_im[d] = make_float2(v0.y, v1.y);
         }
         __syncthreads();
 
@@ -1287,7 +1293,7 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
     if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
     int grid = (int)std::min<long long>(tiles, cap);
     const char* mf = getenv("FSDR_FIR_MFMA");
-    if (f->kk_mfma && mf && atoi(mf) != 0) {
+    if (f->kk_mfma && (!mf || atoi(mf) != 0)) {
         unsigned elems = MFIR_TILE + f->kk_mfma + 8;
         size_t lds = (2 * (size_t)((elems + 31u) & ~31u) + f->kk_mfma + 20)
                      * sizeof(float);
