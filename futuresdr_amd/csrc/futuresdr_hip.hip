@@ -445,7 +445,7 @@ __device__ __forceinline__ unsigned mfma_swz(unsigned idx) {
 }
 
 #define MFIR_BLOCK 256
-#define MFIR_TILE 2048 /* 4 waves x 2 C-tiles x 256 outputs */
+#define MFIR_TILE 1024 /* 4 waves x 256 outputs */
 
 template <int KK>
 __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
@@ -494,35 +494,24 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
         }
         __syncthreads();
 
-        const unsigned ab = (unsigned)wave * 512 + 16u * r16 + k4;
-        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cre1 = cre0;
-        v4f cim0 = cre0, cim1 = cre0;
-        /* two C-tile pairs per wave (outputs wave*512 + {0,256} + tile):
-         * 4 independent accumulator chains fully cover the MFMA issue
-         * rate and halve the staging/barrier overhead per MFMA */
+        const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int s = 0; s < KK / 4; s++) {
-            float a0r = s_re[mfma_swz(ab + 4 * s)];
-            float a0i = s_im[mfma_swz(ab + 4 * s)];
-            float a1r = s_re[mfma_swz(ab + 256 + 4 * s)];
-            float a1i = s_im[mfma_swz(ab + 256 + 4 * s)];
-            cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0r, bfrag[s], cre0,
-                                                        0, 0, 0);
-            cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0i, bfrag[s], cim0,
-                                                        0, 0, 0);
-            cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1r, bfrag[s], cre1,
-                                                        0, 0, 0);
-            cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1i, bfrag[s], cim1,
-                                                        0, 0, 0);
+            float a_re = s_re[mfma_swz(ab + 4 * s)];
+            float a_im = s_im[mfma_swz(ab + 4 * s)];
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
+                                                       0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
+                                                       0, 0, 0);
         }
         /* C layout: col = lane&15, row = (lane>>4)*4 + q (cdna4 16x16) */
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
-            long long o0 = out_base + (long long)wave * 512 + 16 * row + r16;
-            if (o0 < n_out) out[o0] = make_float2(cre0[q], cim0[q]);
-            long long o1 = o0 + 256;
-            if (o1 < n_out) out[o1] = make_float2(cre1[q], cim1[q]);
+            long long o = out_base + (long long)wave * 256 + 16 * row + r16;
+            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
         }
         __syncthreads();
     }
