@@ -477,22 +477,41 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
         bfrag[s] = s_rtx[15 + 4 * s + k4 - r16];
     __syncthreads();
 
+    /* global staging is software-pipelined across the tile loop: tile
+     * t+1's loads are issued right after the staging barrier of tile t,
+     * so their HBM latency hides under t's 72 MFMAs. */
+    constexpr int NL = (MFIR_TILE + KK + 8 + 2 * MFIR_BLOCK - 1) /
+                       (2 * MFIR_BLOCK); /* float4 (2 complex) per slot */
+    float4 stg[NL];
+    auto load_tile = [&](long long tl) {
+        const long long ob = tl * MFIR_TILE;
+#pragma unroll
+        for (int j = 0; j < NL; j++) {
+            unsigned i = 2 * (tid + j * MFIR_BLOCK);
+            long long g = ob + i;
+            float2 v0 = (i < elems && g < n_in_valid)
+                            ? in[g] : make_float2(0.f, 0.f);
+            float2 v1 = (i + 1 < elems && g + 1 < n_in_valid)
+                            ? in[g + 1] : make_float2(0.f, 0.f);
+            stg[j] = make_float4(v0.x, v0.y, v1.x, v1.y);
+        }
+    };
+    load_tile(blockIdx.x);
     for (long long tile = blockIdx.x;
          tile * (long long)MFIR_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * MFIR_TILE;
-        /* 2 complex per thread-iteration: float4 global load, paired b64
-         * LDS writes (the XOR swizzle preserves even-pair adjacency) */
-        for (unsigned i = 2 * tid; i < elems; i += 2 * MFIR_BLOCK) {
-            long long g = out_base + i;
-            float2 v0 = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
-            float2 v1 = (g + 1 < n_in_valid && i + 1 < elems)
-                            ? in[g + 1]
-                            : make_float2(0.f, 0.f);
-            unsigned d = mfma_swz(i);
-            *(float2*)&s_re[d] = make_float2(v0.x, v1.x);
-            *(float2*)&s_im[d] = make_float2(v0.y, v1.y);
+#pragma unroll
+        for (int j = 0; j < NL; j++) {
+            unsigned i = 2 * (tid + j * MFIR_BLOCK);
+            if (i < elems) {
+                unsigned d = mfma_swz(i);
+                *(float2*)&s_re[d] = make_float2(stg[j].x, stg[j].z);
+                *(float2*)&s_im[d] = make_float2(stg[j].y, stg[j].w);
+            }
         }
         __syncthreads();
+        if ((tile + gridDim.x) * (long long)MFIR_TILE < n_out)
+            load_tile(tile + gridDim.x);
 
         const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
         v4f cre = {0.f, 0.f, 0.f, 0.f};
