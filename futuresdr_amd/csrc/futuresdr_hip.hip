@@ -536,6 +536,105 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
     }
 }
 
+/* ---- MFMA phase-split decimating FIR (D=4) --------------------------- *
+ * decimating_fir.rs:80-95 semantics via the phase decomposition
+ * (t = 4u+v => y[k] = sum_v sum_u P_v[k+u]*rtv[v][u], P_v[i] = x[3+v+4i])
+ * with each phase dot-product run as the same implicit GEMM as
+ * k_fir_mfma_tpl: C accumulates over all 4 phases. KKD >= ceil(T/4)+15,
+ * KKD%4==0. Staging is software-pipelined across the tile loop. */
+#define MDFIR_BLOCK 256
+#define MDFIR_TILE 1024 /* decimated outputs per block */
+
+template <int KKD>
+__global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv /* [4][KKD], rtv[v][u] = rt[4u+v] */,
+    long long n_out, long long n_in_valid) {
+    static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
+    const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* planes = (float*)smem;        /* [8][SPm]: re0..3, im0..3 */
+    float* s_rtx = planes + 8u * SPm;    /* [4][KKD+16], 15-zero prologue */
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+
+    for (int i = tid; i < 4 * (KKD + 16); i += MDFIR_BLOCK) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    __syncthreads();
+    float bfrag[4][KKD / 4];
+#pragma unroll
+    for (int v = 0; v < 4; v++)
+#pragma unroll
+        for (int s = 0; s < KKD / 4; s++)
+            bfrag[v][s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+    __syncthreads();
+
+    const unsigned span = 3 + 4 * elemsP; /* input elements per tile */
+    constexpr int NL =
+        (4 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float2 stg[NL];
+    auto load_tile = [&](long long tl) {
+        const long long ib = tl * MDFIR_TILE * 4;
+#pragma unroll
+        for (int j = 0; j < NL; j++) {
+            unsigned rel = 3 + (unsigned)(tid + j * MDFIR_BLOCK);
+            long long g = ib + rel;
+            stg[j] = (rel < span && g < n_in_valid)
+                         ? in[g] : make_float2(0.f, 0.f);
+        }
+    };
+    load_tile(blockIdx.x);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MDFIR_TILE;
+#pragma unroll
+        for (int j = 0; j < NL; j++) {
+            unsigned rel = 3 + (unsigned)(tid + j * MDFIR_BLOCK);
+            if (rel < span) {
+                unsigned v = (rel - 3) & 3u, i = (rel - 3) >> 2;
+                unsigned d = mfma_swz(i);
+                planes[v * SPm + d] = stg[j].x;
+                planes[(4 + v) * SPm + d] = stg[j].y;
+            }
+        }
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_tile(tile + gridDim.x);
+
+        const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int v = 0; v < 4; v++) {
+            const float* pre = planes + (unsigned)v * SPm;
+            const float* pim = planes + (unsigned)(4 + v) * SPm;
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++) {
+                float a_re = pre[mfma_swz(ab + 4 * s)];
+                float a_im = pim[mfma_swz(ab + 4 * s)];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[v][s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[v][s], cim, 0, 0, 0);
+            }
+        }
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            long long o = out_base + (long long)wave * 256 + 16 * row + r16;
+            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+    }
+}
+
 /* ---- Phase-split decimating FIR, D=4, compile-time taps -------------- *
  * Same math as decimating_fir.rs:80-95 (D=4): y[k] = sum_t x[3+4k+t] *
  * h[T-1-t]. Decompose t = 4u+v: y[k] = sum_v sum_u P_v[k+u] * rt[4u+v]
@@ -1180,6 +1279,30 @@ extern "C" fsdr_filter* fsdr_decim_fir_cf32_create(size_t decimation,
         return nullptr;
     }
     if (decimation == 4 && n_taps <= 512) {
+        /* MFMA variant: per-phase K = ceil(T/4)+15 rounded to 4 */
+        static const int kkds[] = {20, 32, 48, 80, 144};
+        size_t tpd_true = (n_taps + 3) / 4;
+        for (int k : kkds)
+            if ((size_t)k >= tpd_true + 15) { f->kk_mfma = k; break; }
+        if (f->kk_mfma) {
+            int kkd = f->kk_mfma;
+            std::vector<float> rm(4 * (size_t)kkd, 0.f);
+            for (int v = 0; v < 4; v++)
+                for (int u = 0; u < kkd; u++) {
+                    size_t t = 4 * (size_t)u + v;
+                    if (t < n_taps)
+                        rm[v * kkd + u] = taps[n_taps - 1 - t];
+                }
+            if (hipMalloc(&f->d_mtaps, rm.size() * sizeof(float)) !=
+                    hipSuccess ||
+                hipMemcpy(f->d_mtaps, rm.data(),
+                          rm.size() * sizeof(float),
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                set_err("decim mfma taps upload failed");
+                delete f;
+                return nullptr;
+            }
+        }
         /* phase-split template: rtv[v][u] = h[n_taps-1-(4u+v)] */
         static const int tpds[] = {8, 16, 32, 64, 128};
         for (int t : tpds) {
@@ -1365,6 +1488,30 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
 static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
                              size_t n_out, size_t n_in, hipStream_t st) {
     if (n_out == 0) return FSDR_OK;
+    const char* dmf = getenv("FSDR_DECIM_MFMA");
+    if (f->decim == 4 && f->kk_mfma && (!dmf || atoi(dmf) != 0)) {
+        long long tiles = ((long long)n_out + MDFIR_TILE - 1) / MDFIR_TILE;
+        long long cap = 256 * 64;
+        if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
+        int grid = (int)std::min<long long>(tiles, cap);
+        unsigned elemsP = MDFIR_TILE + f->kk_mfma + 8;
+        size_t lds = (8 * (size_t)((elemsP + 31u) & ~31u) +
+                      4 * ((size_t)f->kk_mfma + 16)) * sizeof(float);
+#define MDFIR_TPL_CASE(KV)                                                       case KV:                                                                         hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_mfma_tpl<KV>),                                      dim3(grid), dim3(MDFIR_BLOCK), lds, st,                                      (const float2*)d_in, (float2*)d_out, f->d_mtaps,                             (long long)n_out, (long long)n_in);                       break;
+        switch (f->kk_mfma) {
+            MDFIR_TPL_CASE(20)
+            MDFIR_TPL_CASE(32)
+            MDFIR_TPL_CASE(48)
+            MDFIR_TPL_CASE(80)
+            MDFIR_TPL_CASE(144)
+            default:
+                set_err("bad decim mfma K");
+                return FSDR_ERR_INVALID;
+        }
+#undef MDFIR_TPL_CASE
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
+    }
     if (f->decim == 4 && f->tp_tpl) {
         long long tiles = ((long long)n_out + DFIRT_TILE - 1) / DFIRT_TILE;
         long long cap = 256 * 64;
