@@ -568,13 +568,6 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
         s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
     }
     __syncthreads();
-    float bfrag[4][KKD / 4];
-#pragma unroll
-    for (int v = 0; v < 4; v++)
-#pragma unroll
-        for (int s = 0; s < KKD / 4; s++)
-            bfrag[v][s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
-    __syncthreads();
 
     const unsigned span = 3 + 4 * elemsP; /* input elements per tile */
     constexpr int NL =
@@ -615,14 +608,21 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
         for (int v = 0; v < 4; v++) {
             const float* pre = planes + (unsigned)v * SPm;
             const float* pim = planes + (unsigned)(4 + v) * SPm;
+            /* per-phase B fragments re-read from LDS each tile (48
+             * broadcast reads/tile) instead of 48 resident VGPRs — the
+             * register pressure capped occupancy at 2 waves/SIMD */
+            float bfrag[KKD / 4];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++)
+                bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
 #pragma unroll
             for (int s = 0; s < KKD / 4; s++) {
                 float a_re = pre[mfma_swz(ab + 4 * s)];
                 float a_im = pim[mfma_swz(ab + 4 * s)];
                 cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[v][s], cre, 0, 0, 0);
+                    a_re, bfrag[s], cre, 0, 0, 0);
                 cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[v][s], cim, 0, 0, 0);
+                    a_im, bfrag[s], cim, 0, 0, 0);
             }
         }
 #pragma unroll
