@@ -536,6 +536,100 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
     }
 }
 
+/* ---- MFMA FIR, 32x32x2 variant --------------------------------------- *
+ * Same implicit GEMM as k_fir_mfma_tpl but on v_mfma_f32_32x32x2_f32:
+ * one 32x32 C tile pair (re+im) = 1024 outputs per WAVE, single-wave
+ * blocks — 4x less staging and barrier overhead per output than the
+ * 16x16 shape (at the cost of K = T+31 shift inflation). B fragments are
+ * re-read from LDS per tile (80 broadcast reads) to keep registers for
+ * occupancy. A-plane XOR swizzle: idx ^ (((idx>>5)&15)<<1) makes the
+ * 32-dword row stride conflict-free up to a residual 2-way (rows r,
+ * r+16) while preserving even-pair adjacency for the staging writes. */
+__device__ __forceinline__ unsigned mfma32_swz(unsigned idx) {
+    return idx ^ (((idx >> 5) & 15u) << 1);
+}
+
+#define MFIR32_BLOCK 64
+#define MFIR32_TILE 1024 /* one wave, one 32x32 C pair */
+
+typedef float v16f __attribute__((ext_vector_type(16)));
+
+template <int KK2>
+__global__ __launch_bounds__(MFIR32_BLOCK) void k_fir_mfma32_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtaps /* reversed, zero-filled to KK2 */,
+    long long n_out, long long n_in_valid) {
+    static_assert(KK2 % 2 == 0, "KK2 must be even");
+    const unsigned elems = MFIR32_TILE + KK2 + 8;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* s_re = (float*)smem;
+    float* s_im = s_re + ((elems + 31u) & ~31u);
+    float* s_rtx = s_im + ((elems + 31u) & ~31u); /* 31 zeros + rt + 1 */
+
+    const int tid = threadIdx.x;
+    const int r32 = tid & 31; /* A row / C col */
+    const int k2 = tid >> 5;  /* k sub-slice 0..1 */
+
+    for (int i = tid; i < KK2 + 32; i += MFIR32_BLOCK)
+        s_rtx[i] = (i >= 31 && i < 31 + KK2) ? rtaps[i - 31] : 0.f;
+    __syncthreads();
+
+    constexpr int NL =
+        (MFIR32_TILE + KK2 + 8 + 2 * MFIR32_BLOCK - 1) / (2 * MFIR32_BLOCK);
+    float4 stg[NL];
+    auto load_tile = [&](long long tl) {
+        const long long ob = tl * MFIR32_TILE;
+#pragma unroll
+        for (int j = 0; j < NL; j++) {
+            unsigned i = 2 * (tid + j * MFIR32_BLOCK);
+            long long g = ob + i;
+            float2 v0 = (i < elems && g < n_in_valid)
+                            ? in[g] : make_float2(0.f, 0.f);
+            float2 v1 = (i + 1 < elems && g + 1 < n_in_valid)
+                            ? in[g + 1] : make_float2(0.f, 0.f);
+            stg[j] = make_float4(v0.x, v0.y, v1.x, v1.y);
+        }
+    };
+    load_tile(blockIdx.x);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MFIR32_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MFIR32_TILE;
+#pragma unroll
+        for (int j = 0; j < NL; j++) {
+            unsigned i = 2 * (tid + j * MFIR32_BLOCK);
+            if (i < elems) {
+                unsigned d = mfma32_swz(i);
+                *(float2*)&s_re[d] = make_float2(stg[j].x, stg[j].z);
+                *(float2*)&s_im[d] = make_float2(stg[j].y, stg[j].w);
+            }
+        }
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)MFIR32_TILE < n_out)
+            load_tile(tile + gridDim.x);
+
+        const unsigned ab = 32u * r32 + k2;
+        v16f cre = {};
+        v16f cim = {};
+#pragma unroll
+        for (int s = 0; s < KK2 / 2; s++) {
+            float b = s_rtx[31 + 2 * s + k2 - r32];
+            float a_re = s_re[mfma32_swz(ab + 2 * s)];
+            float a_im = s_im[mfma32_swz(ab + 2 * s)];
+            cre = __builtin_amdgcn_mfma_f32_32x32x2f32(a_re, b, cre, 0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_32x32x2f32(a_im, b, cim, 0, 0, 0);
+        }
+        /* C layout (cdna4 32x32): col = lane&31,
+         * row = (reg&3) + 8*(reg>>2) + 4*(lane>>5) */
+#pragma unroll
+        for (int q = 0; q < 16; q++) {
+            int row = (q & 3) + 8 * (q >> 2) + 4 * k2;
+            long long o = out_base + 32 * row + r32;
+            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+    }
+}
+
 /* ---- MFMA phase-split decimating FIR (D=4) --------------------------- *
  * decimating_fir.rs:80-95 semantics via the phase decomposition
  * (t = 4u+v => y[k] = sum_v sum_u P_v[k+u]*rtv[v][u], P_v[i] = x[3+v+4i])
@@ -1139,6 +1233,8 @@ struct fsdr_filter {
     float* d_rtaps = nullptr; /* reversed taps zero-filled to tp_tpl */
     int kk_mfma = 0;         /* MFMA variant K (>= n_taps+15, %4==0) */
     float* d_mtaps = nullptr; /* reversed taps zero-filled to kk_mfma */
+    int kk_mfma32 = 0;       /* 32x32 variant K (>= n_taps+31, even) */
+    float* d_mtaps32 = nullptr;
     float2* d_twid = nullptr;
     /* staging buffers for the host-span path */
     void* d_in = nullptr;
@@ -1205,6 +1301,21 @@ extern "C" fsdr_filter* fsdr_fir_cf32_create(const float* taps,
             hipMemcpy(f->d_rtaps, rt.data(), rt.size() * sizeof(float),
                       hipMemcpyHostToDevice) != hipSuccess) {
             set_err("reversed taps upload failed");
+            delete f;
+            return nullptr;
+        }
+    }
+    static const int kk32s[] = {48, 96, 160, 288, 544};
+    for (int k : kk32s)
+        if ((size_t)k >= n_taps + 31) { f->kk_mfma32 = k; break; }
+    if (f->kk_mfma32) {
+        std::vector<float> rt(f->kk_mfma32, 0.f);
+        for (size_t i = 0; i < n_taps; i++) rt[i] = taps[n_taps - 1 - i];
+        if (hipMalloc(&f->d_mtaps32, rt.size() * sizeof(float)) !=
+                hipSuccess ||
+            hipMemcpy(f->d_mtaps32, rt.data(), rt.size() * sizeof(float),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_err("mfma32 taps upload failed");
             delete f;
             return nullptr;
         }
@@ -1408,6 +1519,7 @@ extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (f->d_taps) (void)hipFree(f->d_taps);
     if (f->d_rtaps) (void)hipFree(f->d_rtaps);
     if (f->d_mtaps) (void)hipFree(f->d_mtaps);
+    if (f->d_mtaps32) (void)hipFree(f->d_mtaps32);
     if (f->d_twid) (void)hipFree(f->d_twid);
     if (f->d_in) (void)hipFree(f->d_in);
     if (f->d_out) (void)hipFree(f->d_out);
@@ -1434,6 +1546,33 @@ static int launch_fir_cf32(fsdr_filter* f, const void* d_in, void* d_out,
     long long cap = 256 * 16;
     if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
     int grid = (int)std::min<long long>(tiles, cap);
+    const char* mf32 = getenv("FSDR_FIR_MFMA32");
+    if (f->kk_mfma && mf32 && atoi(mf32) != 0) {
+        /* 32x32 variant reuses d_mtaps when kk32 <= allocated; host
+         * uploads a separate array sized for K = T+31 */
+        long long tiles32 =
+            ((long long)n_out + MFIR32_TILE - 1) / MFIR32_TILE;
+        long long cap32 = 256 * 64;
+        if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap32 = atoll(e);
+        int grid32 = (int)std::min<long long>(tiles32, cap32);
+        unsigned elems32 = MFIR32_TILE + f->kk_mfma32 + 8;
+        size_t lds32 = (2 * (size_t)((elems32 + 31u) & ~31u) +
+                        f->kk_mfma32 + 36) * sizeof(float);
+#define MFIR32_TPL_CASE(KV)                                                      case KV:                                                                         hipLaunchKernelGGL(HIP_KERNEL_NAME(k_fir_mfma32_tpl<KV>),                                       dim3(grid32), dim3(MFIR32_BLOCK), lds32, st,                                 (const float2*)d_in, (float2*)d_out,                                         f->d_mtaps32, (long long)n_out, (long long)n_in);         break;
+        switch (f->kk_mfma32) {
+            MFIR32_TPL_CASE(48)
+            MFIR32_TPL_CASE(96)
+            MFIR32_TPL_CASE(160)
+            MFIR32_TPL_CASE(288)
+            MFIR32_TPL_CASE(544)
+            default:
+                set_err("bad mfma32 K");
+                return FSDR_ERR_INVALID;
+        }
+#undef MFIR32_TPL_CASE
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
+    }
     const char* mf = getenv("FSDR_FIR_MFMA");
     if (f->kk_mfma && (!mf || atoi(mf) != 0)) {
         unsigned elems = MFIR_TILE + f->kk_mfma + 8;
