@@ -85,6 +85,8 @@ def _load():
     lib.fsdr_fft_cf32_create.restype = vp
     lib.fsdr_fft_cf32_create.argtypes = [sz, ctypes.c_int, ctypes.c_int, f32p]
     lib.fsdr_mag2_create.restype = vp
+    lib.fsdr_moving_avg_create.restype = vp
+    lib.fsdr_moving_avg_create.argtypes = [sz, ctypes.c_float, sz]
     lib.fsdr_filter_length.restype = sz
     lib.fsdr_filter_length.argtypes = [vp]
     lib.fsdr_filter_host.restype = ctypes.c_int
@@ -337,6 +339,17 @@ class Mag2(Filter):
 
     def __init__(self):
         super().__init__(_load().fsdr_mag2_create())
+
+
+class MovingAvg(Filter):
+    """MovingAvg block — moving_avg.rs:79-118 (stateful per-bin EMA)."""
+
+    ITEM_IN = np.dtype(np.float32)
+    ITEM_OUT = np.dtype(np.float32)
+
+    def __init__(self, width, decay_factor, history):
+        super().__init__(_load().fsdr_moving_avg_create(
+            width, decay_factor, history))
 
 
 class Chain:

@@ -914,6 +914,35 @@ __global__ __launch_bounds__(256) void k_fir_ccf32(
     }
 }
 
+/* ================= MovingAvg ========================================== *
+ * src/blocks/moving_avg.rs:79-118: per-bin EMA over WIDTH-sized frames,
+ * emit every `history` frames; avg state lives in HBM (stateful block).
+ * One lane per bin, sequential over frames (frames are a recurrence). */
+__global__ void k_moving_avg(const float* __restrict__ in,
+                             float* __restrict__ out,
+                             float* __restrict__ avg, int width,
+                             long long frames, int i0, int history,
+                             float decay) {
+    int b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= width) return;
+    float a = avg[b];
+    int i = i0;
+    long long prod = 0;
+    for (long long f = 0; f < frames; f++) {
+        float t = in[f * width + b];
+        if (isfinite(t))
+            a = (1.0f - decay) * a + decay * t;
+        else
+            a *= 1.0f - decay;
+        if (++i == history) {
+            out[prod * width + b] = a;
+            i = 0;
+            prod++;
+        }
+    }
+    avg[b] = a;
+}
+
 /* ================= Rotator ============================================ *
  * Rotator::rotate — crates/futuredsp/src/rotator.rs:23-49: out[i] =
  * in[i] * phase0 * e^{i*theta*(i+1)}. The reference iterates
@@ -1218,10 +1247,15 @@ static fsdr_filter_result resamp_status(size_t L, size_t M, size_t nt_total,
 /* ================= filter handles ===================================== */
 
 enum FilterKind { K_FIR_CF32, K_FIR_F32, K_DECIM_CF32, K_RESAMP_CF32,
-                  K_FFT_CF32, K_MAG2, K_FIR_CCF32 };
+                  K_FFT_CF32, K_MAG2, K_FIR_CCF32, K_MOVAVG };
 
 struct fsdr_filter {
     FilterKind kind;
+    size_t width = 0;        /* MovingAvg */
+    size_t history = 0;
+    size_t i_state = 0;
+    float decay = 0.f;
+    float* d_avg = nullptr;
     size_t n_taps = 0;       /* true tap count (length()) */
     size_t decim = 1, interp = 1;
     size_t fft_len = 0;
@@ -1502,6 +1536,31 @@ extern "C" fsdr_filter* fsdr_mag2_create(void) {
     return f;
 }
 
+extern "C" fsdr_filter* fsdr_moving_avg_create(size_t width,
+                                               float decay_factor,
+                                               size_t history) {
+    if (width == 0 || history == 0 || !(decay_factor >= 0.f) ||
+        decay_factor > 1.f) {
+        /* moving_avg.rs:59-62 assert */
+        set_err("decay_factor must be in [0,1], width/history > 0");
+        return nullptr;
+    }
+    fsdr_filter* f = create_common(K_MOVAVG);
+    if (!f) return nullptr;
+    f->width = width;
+    f->history = history;
+    f->decay = decay_factor;
+    f->item_in = f->item_out = 4;
+    f->n_taps = width; /* length() = min_items analogue */
+    if (hipMalloc(&f->d_avg, width * sizeof(float)) != hipSuccess ||
+        hipMemset(f->d_avg, 0, width * sizeof(float)) != hipSuccess) {
+        set_err("avg state alloc failed");
+        delete f;
+        return nullptr;
+    }
+    return f;
+}
+
 extern "C" size_t fsdr_filter_length(const fsdr_filter* f) {
     return f ? (f->kind == K_MAG2 ? 1 : f->n_taps) : 0;
 }
@@ -1516,6 +1575,7 @@ extern "C" size_t fsdr_filter_item_sizes(const fsdr_filter* f,
 
 extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (!f) return;
+    if (f->d_avg) (void)hipFree(f->d_avg);
     if (f->d_taps) (void)hipFree(f->d_taps);
     if (f->d_rtaps) (void)hipFree(f->d_rtaps);
     if (f->d_mtaps) (void)hipFree(f->d_mtaps);
@@ -1782,6 +1842,29 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                                (const float2*)f->d_taps, (int)f->n_taps,
                                (long long)r->produced, (long long)n_in);
             HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
+        case K_MOVAVG: {
+            /* replicate the work() counting loop on the host */
+            size_t in_frames = n_in / f->width;
+            size_t out_frames = n_out / f->width;
+            size_t cons = 0, prod = 0, i = f->i_state;
+            while (cons < in_frames && prod < out_frames) {
+                if (++i == f->history) { i = 0; prod++; }
+                cons++;
+            }
+            r->consumed = cons * f->width;
+            r->produced = prod * f->width;
+            r->status = FSDR_BOTH_SUFFICIENT;
+            if (cons == 0) return FSDR_OK;
+            hipLaunchKernelGGL(k_moving_avg,
+                               dim3((unsigned)((f->width + 255) / 256)),
+                               dim3(256), 0, st, (const float*)d_in,
+                               (float*)d_out, f->d_avg, (int)f->width,
+                               (long long)cons, (int)f->i_state,
+                               (int)f->history, f->decay);
+            HIP_TRY(hipGetLastError());
+            f->i_state = i;
             return FSDR_OK;
         }
         case K_MAG2: {

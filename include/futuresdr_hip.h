@@ -104,6 +104,10 @@ fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse, int fft_shift,
                                   const float* normalize);
 /* Apply |x|^2 (Complex32 -> f32) — the spectrum mag^2 map. */
 fsdr_filter* fsdr_mag2_create(void);
+/* MovingAvg block (src/blocks/moving_avg.rs:79-118): stateful per-bin
+ * EMA over width-sized f32 frames, emitting every `history` frames. */
+fsdr_filter* fsdr_moving_avg_create(size_t width, float decay_factor,
+                                    size_t history);
 
 /* Filter::length() — lib.rs:65-67. */
 size_t fsdr_filter_length(const fsdr_filter* f);

@@ -45,7 +45,7 @@ def _load():
     if not os.path.exists(_SO):
         build()
     lib = ctypes.CDLL(_SO)
-    szp = ctypes.POINTER(ctypes.c_size_t)  # noqa: F841
+    szp = ctypes.POINTER(ctypes.c_size_t)
     f32p = ctypes.POINTER(ctypes.c_float)
     f64p = ctypes.POINTER(ctypes.c_double)
     vp = ctypes.c_void_p
@@ -88,6 +88,9 @@ def _load():
     lib.oracle_kaiser_multirate_f32.restype = sz
     lib.oracle_kaiser_multirate_f32.argtypes = [
         sz, sz, sz, ctypes.c_double, f32p, sz]
+    lib.oracle_moving_avg.restype = None
+    lib.oracle_moving_avg.argtypes = [sz, ctypes.c_float, sz, f32p, szp,
+                                      vp, sz, vp, sz, szp, szp]
     lib.oracle_chain_cf32.restype = sz
     lib.oracle_chain_cf32.argtypes = [f32p, sz, f32p, sz, sz, sz,
                                       vp, sz, vp, sz, ctypes.c_int]
@@ -263,6 +266,22 @@ def kaiser_multirate_f32(interp, decim, half_polyphase_len, max_ripple):
     lib.oracle_kaiser_multirate_f32(interp, decim, half_polyphase_len,
                                     max_ripple, _f32p(out), out.size)
     return out
+
+
+def moving_avg(width, decay, history, inp, n_out, avg=None, i_state=0):
+    """MovingAvg work() — moving_avg.rs:79-118. Returns
+    (out, consumed, produced, avg_state, i_state)."""
+    lib = _load()
+    inp = np.ascontiguousarray(inp, np.float32)
+    out = np.zeros(n_out, np.float32)
+    avg = np.zeros(width, np.float32) if avg is None else         np.ascontiguousarray(avg, np.float32)
+    ist = ctypes.c_size_t(i_state)
+    cons = ctypes.c_size_t()
+    prod = ctypes.c_size_t()
+    lib.oracle_moving_avg(width, decay, history, _f32p(avg),
+                          ctypes.byref(ist), _c(inp), inp.size, _c(out),
+                          out.size, ctypes.byref(cons), ctypes.byref(prod))
+    return out[:prod.value], cons.value, prod.value, avg, ist.value
 
 
 def chain_cf32(taps1, taps2, decim, fft_len, inp, capture=True, nthreads=0):

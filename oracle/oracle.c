@@ -450,6 +450,34 @@ size_t oracle_kaiser_multirate_f32(size_t interp, size_t decim,
     return num_taps;
 }
 
+/* moving_avg.rs:79-118 */
+void oracle_moving_avg(size_t width, float decay_factor, size_t history,
+                       float* avg, size_t* i_state,
+                       const float* in, size_t n_in,
+                       float* out, size_t n_out,
+                       size_t* consumed, size_t* produced) {
+    size_t cons = 0, prod = 0, i = *i_state;
+    while ((cons + 1) * width <= n_in && (prod + 1) * width <= n_out) {
+        for (size_t b = 0; b < width; b++) {
+            float t = in[cons * width + b];
+            if (isfinite(t))
+                avg[b] = (1.0f - decay_factor) * avg[b] + decay_factor * t;
+            else
+                avg[b] *= 1.0f - decay_factor;
+        }
+        i++;
+        if (i == history) {
+            memcpy(out + prod * width, avg, width * sizeof(float));
+            i = 0;
+            prod++;
+        }
+        cons++;
+    }
+    *i_state = i;
+    *consumed = cons * width;
+    *produced = prod * width;
+}
+
 /* ---------- CPU-baseline chain ---------------------------------------- */
 
 /* f32 iterative radix-2 FFT (forward, unnormalized) for the baseline leg:

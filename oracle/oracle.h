@@ -129,6 +129,17 @@ size_t oracle_kaiser_multirate_f32(size_t interp, size_t decim,
                                    size_t half_polyphase_len,
                                    double max_ripple, float* out, size_t cap);
 
+/* MovingAvg block work() — src/blocks/moving_avg.rs:79-118: per-bin EMA
+ * over WIDTH-sized frames, emitting one averaged frame every
+ * history_size inputs; avg[] and *i_state are carried state. Consumes
+ * while input has a full frame AND output has room for one more frame
+ * (the reference's while condition). */
+void oracle_moving_avg(size_t width, float decay_factor, size_t history,
+                       float* avg, size_t* i_state,
+                       const float* in, size_t n_in,
+                       float* out, size_t n_out,
+                       size_t* consumed, size_t* produced);
+
 /* CPU-baseline chain: FIR(taps1) → decim-by-D(taps2) → per-frame
  * unnormalized fft_len-pt forward DFT (f32 radix-2, same work as the GPU
  * chain), OpenMP-sharded over contiguous chunks with (n_taps-1)-sample

@@ -95,3 +95,62 @@ def test_fg_small_buffer_many_chunks(gpu, oracle_lib):
     ref, _, p, _ = oracle_lib.fir_cf32(taps, x, x.size)
     err = np.abs(got - ref).max() / max(1.0, np.abs(ref).max())
     assert err < 1e-5
+
+
+def test_fg_spectrum_sink_with_moving_avg(gpu, oracle_lib):
+    """Full spectrum pipeline incl. the averaged sink:
+    VectorSource -> Fir -> DecimFir(4) -> Fft(64) -> Mag2 ->
+    MovingAvg(64, 0.1, 2) -> VectorSink (examples/spectrum shape with
+    moving_avg.rs smoothing)."""
+    r = rng(5)
+    x = cplx(r, 4 * 64 * 9 + 600)
+    t1 = r.uniform(-1, 1, 63).astype(np.float32)
+    t2 = r.uniform(-1, 1, 63).astype(np.float32)
+    fg = gpu.Flowgraph()
+    src = fg.vector_source(x)
+    f1 = fg.filter(gpu.Fir(t1))
+    f2 = fg.filter(gpu.DecimFir(4, t2))
+    f3 = fg.filter(gpu.Fft(64))
+    f4 = fg.filter(gpu.Mag2())
+    f5 = fg.filter(gpu.MovingAvg(64, 0.1, 2))
+    snk = fg.vector_sink()
+    fg.connect(src, f1, f2, f3, f4, f5, snk)
+    fg.run()
+    got = fg.sink_data(snk, np.float32)
+    import oracle as o
+    y1, _, _, _ = o.fir_cf32(t1, x, x.size)
+    y2, _, _, _ = o.decim_fir_cf32(4, t2, y1, y1.size)
+    frames = y2.size // 64
+    spec = np.concatenate([o.dft_cf32(y2[i * 64:(i + 1) * 64])
+                           for i in range(frames)])
+    mags = o.mag2(spec)
+    ref, _, _, _, _ = o.moving_avg(64, 0.1, 2, mags, mags.size)
+    assert got.size == ref.size
+    assert_close(got, ref, 1e-4)
+
+
+def test_fg_config3_resampler_chain(gpu, oracle_lib):
+    """BASELINE configs[2] shape with the polyphase resampler:
+    VectorSource -> Fir -> Resampler(1,4) -> Fft(256) -> VectorSink."""
+    r = rng(6)
+    x = cplx(r, 4 * 256 * 5 + 800)
+    t1 = r.uniform(-1, 1, 127).astype(np.float32)
+    t2 = r.uniform(-1, 1, 128).astype(np.float32)
+    fg = gpu.Flowgraph()
+    src = fg.vector_source(x)
+    f1 = fg.filter(gpu.Fir(t1))
+    f2 = fg.filter(gpu.Resampler(1, 4, t2))
+    f3 = fg.filter(gpu.Fft(256))
+    snk = fg.vector_sink()
+    fg.connect(src, f1, f2, f3, snk)
+    fg.run()
+    got = fg.sink_data(snk)
+    import oracle as o
+    y1, _, _, _ = o.fir_cf32(t1, x, x.size)
+    y2, _, _, _ = o.resamp_cf32(1, 4, t2, y1, y1.size)
+    frames = y2.size // 256
+    ref = np.concatenate([o.dft_cf32(y2[i * 256:(i + 1) * 256])
+                          for i in range(frames)])
+    assert got.size == ref.size
+    rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+    assert rel < 1e-4

@@ -522,3 +522,24 @@ def test_rotator_vs_ideal_and_oracle(gpu, oracle_lib):
     ref, ref_phase = oracle_lib.rotator(theta, x)
     assert np.abs(got - ref).max() < 5e-4
     assert abs(final - cmath.exp(1j * theta * n)) < 1e-5
+
+
+# ---------------- MovingAvg --------------------------------------------
+
+def test_moving_avg_parity(gpu, oracle_lib):
+    r = rng(97)
+    w, d, h = 256, 0.1, 3
+    frames = 32
+    x = r.uniform(-1, 1, frames * w).astype(np.float32)
+    f = gpu.MovingAvg(w, d, h)
+    got, c, p, s = f.filter(x, frames * w)
+    ref, co, po, avg, i = oracle_lib.moving_avg(w, d, h, x, frames * w)
+    assert (c, p) == (co, po)
+    assert_close(got, ref, 1e-6)
+    # state carries across calls (block is stateful)
+    x2 = r.uniform(-1, 1, 5 * w).astype(np.float32)
+    got2, c2, p2, s2 = f.filter(x2, 5 * w)
+    ref2, co2, po2, avg, i = oracle_lib.moving_avg(w, d, h, x2, 5 * w,
+                                                   avg=avg, i_state=i)
+    assert (c2, p2) == (co2, po2)
+    assert_close(got2, ref2, 1e-6)

@@ -242,3 +242,29 @@ def test_chain_matches_composition(oracle_lib):
     # within fp tolerance; per-frame computation is deterministic)
     ch4, _ = o.chain_cf32(t1, t2, 4, 1024, inp, nthreads=4)
     np.testing.assert_array_equal(ch, ch4)
+
+
+def test_moving_avg_semantics(oracle_lib):
+    # moving_avg.rs:92-118: EMA update + emit every history frames;
+    # non-finite inputs only decay the average (:95-99)
+    o = oracle_lib
+    w, d, h = 4, 0.25, 2
+    x = np.arange(4 * w, dtype=np.float32)
+    out, cons, prod, avg, i = o.moving_avg(w, d, h, x, 4 * w)
+    assert cons == 4 * w and prod == 2 * w and i == 0
+    # manual EMA
+    a = np.zeros(w)
+    emitted = []
+    for f in range(4):
+        a = 0.75 * a + 0.25 * x[f * w:(f + 1) * w]
+        if (f + 1) % h == 0:
+            emitted.append(a.copy())
+    np.testing.assert_allclose(out, np.concatenate(emitted), rtol=1e-6)
+    # output-limited: consumption stops when no room for the next frame
+    out, cons, prod, avg, i = o.moving_avg(w, d, h, x, w)
+    assert prod == w and cons == 2 * w
+    # nan input decays only
+    xn = np.full(w, np.nan, np.float32)
+    out, cons, prod, avg2, i = o.moving_avg(w, d, 1, xn, w,
+                                            avg=np.ones(w, np.float32))
+    np.testing.assert_allclose(out, 0.75 * np.ones(w), rtol=1e-6)
