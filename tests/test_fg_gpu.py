@@ -16,6 +16,13 @@ def cplx(r, n):
     return (r.uniform(-1, 1, (n, 2)) @ [1, 1j]).astype(np.complex64)
 
 
+def assert_close(got, ref, tol=1e-5):
+    got, ref = np.asarray(got), np.asarray(ref)
+    scale = max(1.0, float(np.abs(ref).max()) if ref.size else 1.0)
+    err = float(np.abs(got - ref).max()) if ref.size else 0.0
+    assert err <= tol * scale, f"max err {err} > {tol}*{scale}"
+
+
 def test_fg_perf_fir_shape(gpu):
     """NullSource -> Head(samples) -> Fir(64 taps) x stages -> NullSink:
     the perf/fir flowgraph (perf/fir/fir.rs:50-74) with its correctness
