@@ -86,6 +86,10 @@ def _load():
     lib.fsdr_fft_cf32_create.argtypes = [sz, ctypes.c_int, ctypes.c_int, f32p]
     lib.fsdr_mag2_create.restype = vp
     lib.fsdr_moving_avg_create.restype = vp
+    lib.fsdr_xlating_fir_cf32_create.restype = vp
+    lib.fsdr_xlating_fir_cf32_create.argtypes = [f32p, sz, sz,
+                                                 ctypes.c_float,
+                                                 ctypes.c_float]
     lib.fsdr_moving_avg_create.argtypes = [sz, ctypes.c_float, sz]
     lib.fsdr_filter_length.restype = sz
     lib.fsdr_filter_length.argtypes = [vp]
@@ -339,6 +343,15 @@ class Mag2(Filter):
 
     def __init__(self):
         super().__init__(_load().fsdr_mag2_create())
+
+
+class XlatingFir(Filter):
+    """XlatingFir block — xlating_fir.rs (rotate + filter + decimate)."""
+
+    def __init__(self, taps, decimation, offset, sample_rate):
+        self._taps_keep, p = _f32(taps)
+        super().__init__(_load().fsdr_xlating_fir_cf32_create(
+            p, self._taps_keep.size, decimation, offset, sample_rate))
 
 
 class MovingAvg(Filter):

@@ -914,6 +914,44 @@ __global__ __launch_bounds__(256) void k_fir_ccf32(
     }
 }
 
+/* ================= XlatingFir ========================================= *
+ * src/blocks/xlating_fir.rs: DecimatingFir with complex band-pass taps
+ * (bpf[i] = e^{i*TAU*offset/fs * i} * taps[i], :79-88) fused with the
+ * output Rotator (:91-94,116: phase_incr = -TAU*offset*D/fs applied
+ * in-place to the produced samples). Correctness-first kernel (LDS tile,
+ * lane-strided outputs); the rotator phase uses the closed form (the
+ * reference iterates — tolerance note in tests). */
+__global__ __launch_bounds__(256) void k_xlating_decim_ccf32(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float2* __restrict__ taps /* reversed bpf taps */, int n_taps,
+    long long decim, long long n_out, long long n_in_valid, float theta,
+    float p0r, float p0i) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float2* s_h = (float2*)smem;
+    for (int i = threadIdx.x; i < n_taps; i += blockDim.x) s_h[i] = taps[i];
+    __syncthreads();
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long k = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         k < n_out; k += stride) {
+        float sre = 0.f, sim = 0.f;
+        long long base = decim - 1 + k * decim;
+        for (int t = 0; t < n_taps; t++) {
+            float2 x = (base + t < n_in_valid) ? in[base + t]
+                                               : make_float2(0.f, 0.f);
+            float2 hh = s_h[t];
+            sre = fmaf(x.x, hh.x, sre);
+            sre = fmaf(-x.y, hh.y, sre);
+            sim = fmaf(x.x, hh.y, sim);
+            sim = fmaf(x.y, hh.x, sim);
+        }
+        float s, cth;
+        __sincosf(theta * (float)(k + 1), &s, &cth);
+        float pr = cth * p0r - s * p0i;
+        float pi = cth * p0i + s * p0r;
+        out[k] = make_float2(sre * pr - sim * pi, sre * pi + sim * pr);
+    }
+}
+
 /* ================= MovingAvg ========================================== *
  * src/blocks/moving_avg.rs:79-118: per-bin EMA over WIDTH-sized frames,
  * emit every `history` frames; avg state lives in HBM (stateful block).
@@ -1247,7 +1285,8 @@ static fsdr_filter_result resamp_status(size_t L, size_t M, size_t nt_total,
 /* ================= filter handles ===================================== */
 
 enum FilterKind { K_FIR_CF32, K_FIR_F32, K_DECIM_CF32, K_RESAMP_CF32,
-                  K_FFT_CF32, K_MAG2, K_FIR_CCF32, K_MOVAVG };
+                  K_FFT_CF32, K_MAG2, K_FIR_CCF32, K_MOVAVG,
+                  K_XLATING };
 
 struct fsdr_filter {
     FilterKind kind;
@@ -1255,6 +1294,7 @@ struct fsdr_filter {
     size_t history = 0;
     size_t i_state = 0;
     float decay = 0.f;
+    float rot_re = 1.f, rot_im = 0.f; /* xlating rotator phase state */
     float* d_avg = nullptr;
     size_t n_taps = 0;       /* true tap count (length()) */
     size_t decim = 1, interp = 1;
@@ -1533,6 +1573,46 @@ extern "C" fsdr_filter* fsdr_mag2_create(void) {
     fsdr_filter* f = create_common(K_MAG2);
     if (!f) return nullptr;
     f->item_out = 4;
+    return f;
+}
+
+extern "C" fsdr_filter* fsdr_xlating_fir_cf32_create(const float* taps,
+                                                     size_t n_taps,
+                                                     size_t decimation,
+                                                     float offset,
+                                                     float sample_rate) {
+    /* xlating_fir.rs:44 assert decimation >= 2 */
+    if (!taps || n_taps == 0 || decimation < 2) {
+        set_err("xlating fir: taps required, decimation must be >= 2");
+        return nullptr;
+    }
+    fsdr_filter* f = create_common(K_XLATING);
+    if (!f) return nullptr;
+    f->n_taps = n_taps;
+    f->decim = decimation;
+    /* bpf taps, f32 math identical to from_polar (:79-88), stored
+     * REVERSED for the kernel */
+    std::vector<float2> bpf(n_taps);
+    for (size_t i = 0; i < n_taps; i++) {
+        float ang = (float)i * 6.2831853071795864769f * offset / sample_rate;
+        float2 rot = make_float2(cosf(ang), sinf(ang));
+        size_t src_i = i;
+        bpf[n_taps - 1 - src_i] =
+            make_float2(rot.x * taps[src_i], rot.y * taps[src_i]);
+    }
+    if (hipMalloc(&f->d_taps, bpf.size() * sizeof(float2)) != hipSuccess ||
+        hipMemcpy(f->d_taps, bpf.data(), bpf.size() * sizeof(float2),
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        set_err("bpf taps upload failed");
+        delete f;
+        return nullptr;
+    }
+    /* rotator increment (:91-94) and unit start phase (rotator.rs:17-19) */
+    f->decay = -6.2831853071795864769f * offset * (float)decimation /
+               sample_rate;                     /* reuse field as theta */
+    f->rot_re = 1.0f;
+    f->rot_im = 0.0f;
+    f->n_taps_padded = (int)n_taps;
     return f;
 }
 
@@ -1842,6 +1922,28 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                                (const float2*)f->d_taps, (int)f->n_taps,
                                (long long)r->produced, (long long)n_in);
             HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
+        case K_XLATING: {
+            *r = decim_status(f->decim, n_in, f->n_taps, n_out);
+            if (r->produced == 0) return FSDR_OK;
+            size_t lds = f->n_taps * sizeof(float2);
+            hipLaunchKernelGGL(k_xlating_decim_ccf32,
+                               dim3(grid_for((long long)r->produced, 256)),
+                               dim3(256), lds, st, (const float2*)d_in,
+                               (float2*)d_out, (const float2*)f->d_taps,
+                               (int)f->n_taps, (long long)f->decim,
+                               (long long)r->produced, (long long)n_in,
+                               f->decay, f->rot_re, f->rot_im);
+            HIP_TRY(hipGetLastError());
+            { /* advance the rotator phase (closed form, f64) */
+                double a = (double)f->decay * (double)r->produced;
+                double cs = cos(a), sn = sin(a);
+                float nr = (float)(cs * f->rot_re - sn * f->rot_im);
+                float ni = (float)(cs * f->rot_im + sn * f->rot_re);
+                f->rot_re = nr;
+                f->rot_im = ni;
+            }
             return FSDR_OK;
         }
         case K_MOVAVG: {

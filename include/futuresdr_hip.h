@@ -104,6 +104,14 @@ fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse, int fft_shift,
                                   const float* normalize);
 /* Apply |x|^2 (Complex32 -> f32) — the spectrum mag^2 map. */
 fsdr_filter* fsdr_mag2_create(void);
+/* XlatingFir (src/blocks/xlating_fir.rs): rotate-by-offset + filter +
+ * decimate, fused: DecimatingFir with complex band-pass taps
+ * (bpf[i] = e^{i*TAU*offset/fs*i}*taps[i]) and the output Rotator
+ * (phase_incr = -TAU*offset*decimation/fs); rotator phase is carried
+ * state. decimation must be >= 2 (xlating_fir.rs:44). */
+fsdr_filter* fsdr_xlating_fir_cf32_create(const float* taps, size_t n_taps,
+                                          size_t decimation, float offset,
+                                          float sample_rate);
 /* MovingAvg block (src/blocks/moving_avg.rs:79-118): stateful per-bin
  * EMA over width-sized f32 frames, emitting every `history` frames. */
 fsdr_filter* fsdr_moving_avg_create(size_t width, float decay_factor,

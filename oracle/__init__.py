@@ -61,6 +61,8 @@ def _load():
     lib.oracle_decim_fir_f32.argtypes = [sz, f32p, sz, vp, sz, vp, sz]
     lib.oracle_decim_fir_cf32.restype = _Result
     lib.oracle_decim_fir_cf32.argtypes = [sz, f32p, sz, vp, sz, vp, sz]
+    lib.oracle_decim_fir_ccf32.restype = _Result
+    lib.oracle_decim_fir_ccf32.argtypes = [sz, vp, sz, vp, sz, vp, sz]
     lib.oracle_resamp_f32.restype = _Result
     lib.oracle_resamp_f32.argtypes = [sz, sz, f32p, sz, vp, sz, vp, sz]
     lib.oracle_resamp_cf32.restype = _Result
@@ -153,6 +155,16 @@ def decim_fir_cf32(decim, taps, inp, n_out):
     out = np.zeros(n_out, CF32)
     r = lib.oracle_decim_fir_cf32(decim, _f32p(taps), taps.size, _c(inp),
                                   inp.size, _c(out), out.size)
+    return out[: r.produced], r.consumed, r.produced, r.status
+
+
+def decim_fir_ccf32(decim, taps, inp, n_out):
+    lib = _load()
+    taps = np.ascontiguousarray(taps, CF32)
+    inp = np.ascontiguousarray(inp, CF32)
+    out = np.zeros(n_out, CF32)
+    r = lib.oracle_decim_fir_ccf32(decim, _c(taps), taps.size, _c(inp),
+                                   inp.size, _c(out), out.size)
     return out[: r.produced], r.consumed, r.produced, r.status
 
 

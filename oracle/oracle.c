@@ -144,6 +144,26 @@ oracle_result oracle_decim_fir_cf32(size_t decimation,
     return r;
 }
 
+oracle_result oracle_decim_fir_ccf32(size_t decimation,
+                                     const ocf32* taps, size_t n_taps,
+                                     const ocf32* in, size_t n_in,
+                                     ocf32* out, size_t n_out) {
+    /* decimating_fir.rs:183-216 (complex taps, num_complex mul) */
+    oracle_result r = decim_status(decimation, n_in, n_taps, n_out);
+    for (size_t k = 0; k < r.produced; k++) {
+        float sre = 0.0f, sim = 0.0f;
+        for (size_t t = 0; t < n_taps; t++) {
+            ocf32 s = in[decimation - 1 + k * decimation + t];
+            ocf32 hh = taps[n_taps - 1 - t];
+            sre = sre + (s.re * hh.re - s.im * hh.im);
+            sim = sim + (s.re * hh.im + s.im * hh.re);
+        }
+        out[k].re = sre;
+        out[k].im = sim;
+    }
+    return r;
+}
+
 /* ---------- Polyphase resampler --------------------------------------- */
 
 /* resampling_fir_kernel_core — polyphase_resampling_fir.rs:70-124 */

@@ -68,6 +68,13 @@ oracle_result oracle_decim_fir_cf32(size_t decimation,
                                     const ocf32* in, size_t n_in,
                                     ocf32* out, size_t n_out);
 
+/* DecimatingFirFilter<Complex32,Complex32,Complex32> (complex taps, the
+ * XlatingFir core — src/blocks/xlating_fir.rs:79-96,110-127) */
+oracle_result oracle_decim_fir_ccf32(size_t decimation,
+                                     const ocf32* taps, size_t n_taps,
+                                     const ocf32* in, size_t n_in,
+                                     ocf32* out, size_t n_out);
+
 /* PolyphaseResamplingFir<f32,f32,f32>::filter — polyphase_resampling_fir.rs:70-141 */
 oracle_result oracle_resamp_f32(size_t interp, size_t decim,
                                 const float* taps, size_t n_taps,

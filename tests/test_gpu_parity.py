@@ -543,3 +543,32 @@ def test_moving_avg_parity(gpu, oracle_lib):
                                                    avg=avg, i_state=i)
     assert (c2, p2) == (co2, po2)
     assert_close(got2, ref2, 1e-6)
+
+
+# ---------------- XlatingFir -------------------------------------------
+
+def test_xlating_fir_parity(gpu, oracle_lib):
+    """xlating_fir.rs semantics: bpf-tap decimating FIR + output rotator.
+    The oracle pipeline composes oracle_decim_fir_ccf32 with the
+    reference's iterated rotator; the GPU uses closed-form phases, so the
+    comparison tolerance covers the oracle's own O(n*eps) phase drift."""
+    r = rng(117)
+    decim, offset, fs = 4, 12_000.0, 1_000_000.0
+    taps = r.uniform(-1, 1, 63).astype(np.float32)
+    x = cplx(r, 20000)
+    f = gpu.XlatingFir(taps, decim, offset, fs)
+    got, c, p, s = f.filter(x, 20000)
+    # oracle composition
+    i = np.arange(taps.size, dtype=np.float32)
+    ang = i * np.float32(2 * np.pi) * np.float32(offset) / np.float32(fs)
+    bpf = (np.cos(ang) + 1j * np.sin(ang)).astype(np.complex64) * taps
+    ref, co, po, so = oracle_lib.decim_fir_ccf32(decim, bpf, x, 20000)
+    theta = -2 * np.pi * offset * decim / fs
+    ref_rot, _ = oracle_lib.rotator(theta, ref)
+    assert (c, p, s) == (co, po, so)
+    assert_close(got, ref_rot, 5e-4)
+    # stateful phase across calls
+    got2, c2, p2, s2 = f.filter(x, 20000)
+    ideal2 = ref * np.exp(1j * theta * (np.arange(p) + 1 + p)).astype(
+        np.complex64)
+    assert_close(got2, ideal2, 5e-4)
