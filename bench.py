@@ -30,7 +30,6 @@ REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
 FP32_PEAK_TFLOPS = 157.3  # gfx950 fp32 vector/MFMA peak (MI355X_MICROARCH.md)
-FIR_FLOPS_PER_OUT = 127 * 4  # 127 taps x (2 mul + 2 add) — SURVEY.md §8d
 
 
 def log(msg):
@@ -62,48 +61,54 @@ def alloc_dev(lib, bytes_):
     return p
 
 
-def measure_fir_roofline(fa, torch, d_in, n_samples, taps1, traffic_file):
-    """Dominant-kernel (k_fir_cf32) duration via HIP events on the stream
-    the kernel is launched on (torch's current stream)."""
+def measure_chain_roofline(fa, torch, d_in, n_samples, taps1, taps2,
+                           decim, traffic_file):
+    """Dominant-kernel roofline: the chain's dominant kernel is the FUSED
+    combined-taps decimating FIR (k_decim4_mfma_tpl<80>).
+    Duration via HIP events on the launch stream (torch's current
+    stream); achieved = the kernel's algorithmic flops per launch
+    ((T1+T2-1) taps x 4 flops per decimated output) / duration."""
     lib = fa.lib()
     st = torch.cuda.current_stream()
-    fir = fa.Fir(taps1)
-    produced = n_samples + 1 - taps1.size
-    d_y1 = alloc_dev(lib, n_samples * 8)
+    g = np.convolve(taps1.astype(np.float64),
+                    taps2.astype(np.float64)).astype(np.float32)
+    fused = fa.DecimFir(decim, g)
+    produced = (n_samples + 1 - g.size) // decim
+    d_y2 = alloc_dev(lib, (produced + 8) * 8)
     try:
         for _ in range(3):
-            fir.filter_dev(d_in.value, n_samples, d_y1.value, n_samples,
-                           stream=st.cuda_stream)
+            fused.filter_dev(d_in.value, n_samples, d_y2.value, produced,
+                             stream=st.cuda_stream)
         torch.cuda.synchronize()
         reps = 20
         ev0 = torch.cuda.Event(enable_timing=True)
         ev1 = torch.cuda.Event(enable_timing=True)
         ev0.record(st)
         for _ in range(reps):
-            fir.filter_dev(d_in.value, n_samples, d_y1.value, n_samples,
-                           stream=st.cuda_stream)
+            fused.filter_dev(d_in.value, n_samples, d_y2.value, produced,
+                             stream=st.cuda_stream)
         ev1.record(st)
         torch.cuda.synchronize()
         ms = ev0.elapsed_time(ev1) / reps
     finally:
-        lib.fsdr_dev_free(d_y1)
-    flops = produced * FIR_FLOPS_PER_OUT
+        lib.fsdr_dev_free(d_y2)
+    flops = produced * g.size * 4
     achieved_tf = flops / (ms * 1e-3) / 1e12
     traffic = None
     if traffic_file and os.path.exists(traffic_file):
         with open(traffic_file) as f:
             t = json.load(f)
-        bps = t.get("k_fir_cf32_hbm_bytes_per_sample")
+        bps = t.get("fused_decim_hbm_bytes_per_input_sample")
         if bps is not None:
-            traffic = bps * produced
+            traffic = bps * produced * decim
     return {
-        "bound": "mfma",  # fp32 compute-bound: 508 flops vs 16 B per sample
+        "bound": "mfma",  # fp32 MFMA compute-bound: 1012 flops vs ~34 B/out
         "achieved": round(achieved_tf, 2),
         "peak": FP32_PEAK_TFLOPS,
         "unit": "TFLOP/s",
         "frac": round(achieved_tf / FP32_PEAK_TFLOPS, 4),
         "traffic": traffic,
-        "kernel": "k_fir_cf32",
+        "kernel": "k_decim4_mfma_tpl<80> (fused fir127+decim4-127)",
         "ms_per_launch": round(ms, 4),
     }
 
@@ -212,8 +217,9 @@ def main():
     value = n_gpus * S * args.steps / elapsed / 1e6  # whole-job MSample/s
 
     if rank == 0:
-        roofline = measure_fir_roofline(fa, torch, d_in, S, taps1,
-                                        args.traffic_file)
+        roofline = measure_chain_roofline(fa, torch, d_in, S, taps1,
+                                          taps2, args.decim,
+                                          args.traffic_file)
         cpu_baseline = None
         if n_gpus == 1 and not args.skip_cpu_baseline:
             log("measuring CPU baseline (oracle chain, all cores)...")
