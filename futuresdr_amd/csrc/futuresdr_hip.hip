@@ -648,8 +648,12 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
     const unsigned SPm = (elemsP + 31u) & ~31u;
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    float* planes = (float*)smem;        /* [8][SPm]: re0..3, im0..3 */
-    float* s_rtx = planes + 8u * SPm;    /* [4][KKD+16], 15-zero prologue */
+    /* two phases resident at a time ([re_v0, re_v1, im_v0, im_v1]) —
+     * halves LDS vs all-phase planes, doubling resident blocks/CU;
+     * accumulators carry across the two halves, and each half's global
+     * loads are issued under the other half's MFMAs. */
+    float* planes = (float*)smem;        /* [4][SPm] */
+    float* s_rtx = planes + 4u * SPm;    /* [4][KKD+16], 15-zero prologue */
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -664,47 +668,40 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     __syncthreads();
 
     const unsigned span = 3 + 4 * elemsP; /* input elements per tile */
-    constexpr int NL =
-        (4 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
-    float2 stg[NL];
-    auto load_tile = [&](long long tl) {
+    constexpr int NL2 =
+        (2 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float2 stgA[NL2], stgB[NL2];
+    /* elements of phases {2h, 2h+1}: rel = 3 + 4i + 2h + vloc */
+    auto load_half = [&](long long tl, int h, float2 (&stg)[NL2]) {
         const long long ib = tl * MDFIR_TILE * 4;
 #pragma unroll
-        for (int j = 0; j < NL; j++) {
-            unsigned rel = 3 + (unsigned)(tid + j * MDFIR_BLOCK);
+        for (int j = 0; j < NL2; j++) {
+            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+            unsigned rel = 3 + 4 * (idx >> 1) + 2 * h + (idx & 1u);
             long long g = ib + rel;
             stg[j] = (rel < span && g < n_in_valid)
                          ? in[g] : make_float2(0.f, 0.f);
         }
     };
-    load_tile(blockIdx.x);
-    for (long long tile = blockIdx.x;
-         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
-        const long long out_base = tile * MDFIR_TILE;
+    auto write_half = [&](const float2 (&stg)[NL2]) {
 #pragma unroll
-        for (int j = 0; j < NL; j++) {
-            unsigned rel = 3 + (unsigned)(tid + j * MDFIR_BLOCK);
-            if (rel < span) {
-                unsigned v = (rel - 3) & 3u, i = (rel - 3) >> 2;
+        for (int j = 0; j < NL2; j++) {
+            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+            unsigned i = idx >> 1, vloc = idx & 1u;
+            if (i < elemsP) {
                 unsigned d = mfma_swz(i);
-                planes[v * SPm + d] = stg[j].x;
-                planes[(4 + v) * SPm + d] = stg[j].y;
+                planes[vloc * SPm + d] = stg[j].x;
+                planes[(2 + vloc) * SPm + d] = stg[j].y;
             }
         }
-        __syncthreads();
-        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
-            load_tile(tile + gridDim.x);
-
-        const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-        v4f cre = {0.f, 0.f, 0.f, 0.f};
-        v4f cim = {0.f, 0.f, 0.f, 0.f};
+    };
+    const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
 #pragma unroll
-        for (int v = 0; v < 4; v++) {
-            const float* pre = planes + (unsigned)v * SPm;
-            const float* pim = planes + (unsigned)(4 + v) * SPm;
-            /* per-phase B fragments re-read from LDS each tile (48
-             * broadcast reads/tile) instead of 48 resident VGPRs — the
-             * register pressure capped occupancy at 2 waves/SIMD */
+        for (int vloc = 0; vloc < 2; vloc++) {
+            const float* pre = planes + (unsigned)vloc * SPm;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm;
+            const int v = 2 * h + vloc;
             float bfrag[KKD / 4];
 #pragma unroll
             for (int s = 0; s < KKD / 4; s++)
@@ -719,6 +716,24 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
                     a_im, bfrag[s], cim, 0, 0, 0);
             }
         }
+    };
+
+    load_half(blockIdx.x, 0, stgA);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MDFIR_TILE;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        write_half(stgA);
+        __syncthreads();
+        load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
+        mfma_half(0, cre, cim);
+        __syncthreads();
+        write_half(stgB);
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
+        mfma_half(1, cre, cim);
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
@@ -1774,7 +1789,7 @@ static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
         if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
         int grid = (int)std::min<long long>(tiles, cap);
         unsigned elemsP = MDFIR_TILE + f->kk_mfma + 8;
-        size_t lds = (8 * (size_t)((elemsP + 31u) & ~31u) +
+        size_t lds = (4 * (size_t)((elemsP + 31u) & ~31u) +
                       4 * ((size_t)f->kk_mfma + 16)) * sizeof(float);
 #define MDFIR_TPL_CASE(KV)                                                       case KV:                                                                         hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_mfma_tpl<KV>),                                      dim3(grid), dim3(MDFIR_BLOCK), lds, st,                                      (const float2*)d_in, (float2*)d_out, f->d_mtaps,                             (long long)n_out, (long long)n_in);                       break;
         switch (f->kk_mfma) {
