@@ -1097,6 +1097,15 @@ __device__ __forceinline__ float2 cmul_tw(float2 a, float2 w, int inverse) {
     return cmulf(a, w);
 }
 
+/* LDS index swizzle for the ping/pong buffers: float2 element i sits at
+ * bank (2i) mod 64, so accesses whose lane stride is a multiple of 32
+ * elements (the strided Stockham stages) are up-to-8-way conflicted;
+ * XORing bits 5..7 of the element index into bits 2..4 makes them
+ * conflict-free while keeping contiguous stages conflict-free. */
+__device__ __forceinline__ unsigned fft_swz(unsigned i) {
+    return i ^ (((i >> 5) & 7u) << 2);
+}
+
 __global__ __launch_bounds__(256) void k_fft_stockham(
     const float2* __restrict__ in, float2* __restrict__ out,
     float* __restrict__ mag_out /* nullable */,
@@ -1120,9 +1129,9 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
             const float2* src = in + frame * n;
             if (inverse && fft_shift) {       /* fft.rs:179-185: shift input */
                 for (int i = tf; i < n; i += tpf)
-                    a[i] = src[(i + n / 2) % n];
+                    a[fft_swz(i)] = src[(i + n / 2) % n];
             } else {
-                for (int i = tf; i < n; i += tpf) a[i] = src[i];
+                for (int i = tf; i < n; i += tpf) a[fft_swz(i)] = src[i];
             }
         }
         __syncthreads();
@@ -1135,10 +1144,10 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                     const int p = bf / scur;
                     const int q = bf - p * scur;
                     const int tw = n / ncur; /* twiddle stride */
-                    float2 x0 = a[q + scur * p];
-                    float2 x1 = a[q + scur * (p + m4)];
-                    float2 x2 = a[q + scur * (p + 2 * m4)];
-                    float2 x3 = a[q + scur * (p + 3 * m4)];
+                    float2 x0 = a[fft_swz(q + scur * p)];
+                    float2 x1 = a[fft_swz(q + scur * (p + m4))];
+                    float2 x2 = a[fft_swz(q + scur * (p + 2 * m4))];
+                    float2 x3 = a[fft_swz(q + scur * (p + 3 * m4))];
                     /* DIF radix-4: butterflies first, output twiddles
                      * W^p, W^2p, W^3p on frequencies 1..3 (omega4 = -i
                      * forward, +i inverse) */
@@ -1146,14 +1155,14 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                     float2 o0 = f2_add(x1, x3), o1 = f2_sub(x1, x3);
                     float2 o1r = inverse ? make_float2(-o1.y, o1.x)
                                          : make_float2(o1.y, -o1.x);
-                    b[q + scur * (4 * p + 0)] = f2_add(e0, o0);
-                    b[q + scur * (4 * p + 1)] =
+                    b[fft_swz(q + scur * (4 * p + 0))] = f2_add(e0, o0);
+                    b[fft_swz(q + scur * (4 * p + 1))] =
                         cmul_tw(f2_add(e1, o1r), twid[(size_t)p * tw],
                                 inverse);
-                    b[q + scur * (4 * p + 2)] =
+                    b[fft_swz(q + scur * (4 * p + 2))] =
                         cmul_tw(f2_sub(e0, o0), twid[(size_t)2 * p * tw],
                                 inverse);
-                    b[q + scur * (4 * p + 3)] =
+                    b[fft_swz(q + scur * (4 * p + 3))] =
                         cmul_tw(f2_sub(e1, o1r), twid[(size_t)3 * p * tw],
                                 inverse);
                 }
@@ -1168,10 +1177,10 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                 for (int bf = tf; bf < n / 2; bf += tpf) {
                     const int p = bf / scur;
                     const int q = bf - p * scur;
-                    float2 xa = a[q + scur * p];
-                    float2 xb = a[q + scur * (p + 1)];
-                    b[q + scur * 2 * p] = f2_add(xa, xb);
-                    b[q + scur * (2 * p + 1)] =
+                    float2 xa = a[fft_swz(q + scur * p)];
+                    float2 xb = a[fft_swz(q + scur * (p + 1))];
+                    b[fft_swz(q + scur * 2 * p)] = f2_add(xa, xb);
+                    b[fft_swz(q + scur * (2 * p + 1))] =
                         cmul_tw(f2_sub(xa, xb), twid[(size_t)p * (n / 2)],
                                 inverse);
                 }
@@ -1183,7 +1192,7 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
             float2* dst = out + frame * n;
             const bool shift_out = (!inverse) && fft_shift; /* fft.rs:196-204 */
             for (int i = tf; i < n; i += tpf) {
-                float2 v = a[shift_out ? (i + n / 2) % n : i];
+                float2 v = a[fft_swz(shift_out ? (i + n / 2) % n : i)];
                 if (norm != 0.f) { v.x *= norm; v.y *= norm; }
                 dst[i] = v;
                 if (mag_out) mag_out[frame * n + i] = v.x * v.x + v.y * v.y;
