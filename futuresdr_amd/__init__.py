@@ -86,6 +86,11 @@ def _load():
     lib.fsdr_fft_cf32_create.argtypes = [sz, ctypes.c_int, ctypes.c_int, f32p]
     lib.fsdr_mag2_create.restype = vp
     lib.fsdr_moving_avg_create.restype = vp
+    lib.fsdr_pfb_channelizer_create.restype = vp
+    lib.fsdr_pfb_channelizer_create.argtypes = [sz, f32p, sz, ctypes.c_float]
+    lib.fsdr_pfb_channelizer_run_dev.restype = ctypes.c_int
+    lib.fsdr_pfb_channelizer_run_dev.argtypes = [vp, vp, sz, vp, sz, vp,
+                                                 ctypes.POINTER(sz)]
     lib.fsdr_xlating_fir_cf32_create.restype = vp
     lib.fsdr_xlating_fir_cf32_create.argtypes = [f32p, sz, sz,
                                                  ctypes.c_float,
@@ -352,6 +357,44 @@ class XlatingFir(Filter):
         self._taps_keep, p = _f32(taps)
         super().__init__(_load().fsdr_xlating_fir_cf32_create(
             p, self._taps_keep.size, decimation, offset, sample_rate))
+
+
+class PfbChannelizer(Filter):
+    """PFB channelizer — pfb/channelizer.rs (maximally decimated)."""
+
+    def __init__(self, num_channels, taps, oversample_rate=1.0):
+        self._taps_keep, p = _f32(taps)
+        self.n = num_channels
+        super().__init__(_load().fsdr_pfb_channelizer_create(
+            num_channels, p, self._taps_keep.size, oversample_rate))
+
+    def run(self, inp):
+        """Bulk one-shot from zero state over a host span; returns an
+        array of shape [num_channels, produced]."""
+        lib = _load()
+        inp = np.ascontiguousarray(inp, CF32)
+        tpf = -(-self._taps_keep.size // self.n)
+        cap = max(1, (inp.size - self.n * tpf) // self.n)
+        d_in = ctypes.c_void_p()
+        d_out = ctypes.c_void_p()
+        _check(lib.fsdr_dev_alloc(ctypes.byref(d_in), inp.size * 8))
+        _check(lib.fsdr_dev_alloc(ctypes.byref(d_out), self.n * cap * 8))
+        try:
+            _check(lib.fsdr_memcpy_h2d(d_in,
+                                       ctypes.c_void_p(inp.ctypes.data),
+                                       inp.size * 8))
+            prod = ctypes.c_size_t()
+            _check(lib.fsdr_pfb_channelizer_run_dev(
+                self._h, d_in, inp.size, d_out, cap, None,
+                ctypes.byref(prod)))
+            _check(lib.fsdr_synchronize())
+            out = np.zeros(self.n * cap, CF32)
+            _check(lib.fsdr_memcpy_d2h(ctypes.c_void_p(out.ctypes.data),
+                                       d_out, self.n * cap * 8))
+            return out.reshape(self.n, cap)[:, :prod.value]
+        finally:
+            lib.fsdr_dev_free(d_in)
+            lib.fsdr_dev_free(d_out)
 
 
 class MovingAvg(Filter):

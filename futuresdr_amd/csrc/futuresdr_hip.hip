@@ -967,6 +967,48 @@ __global__ __launch_bounds__(256) void k_xlating_decim_ccf32(
     }
 }
 
+/* ================= PFB channelizer (maximally decimated) ============== *
+ * src/blocks/pfb/channelizer.rs (liquid-dsp scheme), bulk form for
+ * oversample_rate = 1 from zero state: with D == N the round-robin
+ * window algebra collapses to
+ *   fft_buf[k][b] = sum_j x[N*(tpf+k-j) + N-1-b] * part[b][j]
+ * (part = partition_filter_taps, utilities.rs:5-25; prefill consumes
+ * N*tpf samples), followed by an unnormalized inverse FFT of size N per
+ * step and a transpose to channel-major outputs. */
+__global__ void k_pfb_dots(const float2* __restrict__ in,
+                           float2* __restrict__ fb,
+                           const float* __restrict__ part /* [N][tpf] */,
+                           int N, int tpf, long long steps) {
+    long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long total = steps * N;
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (; id < total; id += stride) {
+        long long k = id / N;
+        int b = (int)(id - k * N);
+        float sre = 0.f, sim = 0.f;
+        for (int j = 0; j < tpf; j++) {
+            float2 x = in[(long long)N * (tpf + k - j) + (N - 1 - b)];
+            float tap = part[b * tpf + j];
+            sre = fmaf(x.x, tap, sre);
+            sim = fmaf(x.y, tap, sim);
+        }
+        fb[k * N + b] = make_float2(sre, sim);
+    }
+}
+
+__global__ void k_pfb_scatter(const float2* __restrict__ ifft,
+                              float2* __restrict__ out, int N,
+                              long long steps, long long cap) {
+    long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long total = steps * N;
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (; id < total; id += stride) {
+        long long k = id / N;
+        int cidx = (int)(id - k * N);
+        out[cidx * cap + k] = ifft[k * N + cidx];
+    }
+}
+
 /* ================= MovingAvg ========================================== *
  * src/blocks/moving_avg.rs:79-118: per-bin EMA over WIDTH-sized frames,
  * emit every `history` frames; avg state lives in HBM (stateful block).
@@ -1310,7 +1352,7 @@ static fsdr_filter_result resamp_status(size_t L, size_t M, size_t nt_total,
 
 enum FilterKind { K_FIR_CF32, K_FIR_F32, K_DECIM_CF32, K_RESAMP_CF32,
                   K_FFT_CF32, K_MAG2, K_FIR_CCF32, K_MOVAVG,
-                  K_XLATING };
+                  K_XLATING, K_PFB };
 
 struct fsdr_filter {
     FilterKind kind;
@@ -1319,6 +1361,7 @@ struct fsdr_filter {
     size_t i_state = 0;
     float decay = 0.f;
     float rot_re = 1.f, rot_im = 0.f; /* xlating rotator phase state */
+    fsdr_filter* sub = nullptr;       /* channelizer's internal IFFT */
     float* d_avg = nullptr;
     size_t n_taps = 0;       /* true tap count (length()) */
     size_t decim = 1, interp = 1;
@@ -1565,8 +1608,8 @@ extern "C" fsdr_filter* fsdr_resamp_cf32_create(size_t interp, size_t decim,
 extern "C" fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse,
                                              int fft_shift,
                                              const float* normalize) {
-    if (len < 16 || len > 4096 || (len & (len - 1)) != 0) {
-        set_err("fft len must be a power of two in [16,4096]");
+    if (len < 4 || len > 4096 || (len & (len - 1)) != 0) {
+        set_err("fft len must be a power of two in [4,4096]");
         return nullptr;
     }
     fsdr_filter* f = create_common(K_FFT_CF32);
@@ -1640,6 +1683,57 @@ extern "C" fsdr_filter* fsdr_xlating_fir_cf32_create(const float* taps,
     return f;
 }
 
+extern "C" fsdr_filter* fsdr_pfb_channelizer_create(size_t num_channels,
+                                                    const float* taps,
+                                                    size_t n_taps,
+                                                    float oversample_rate) {
+    /* channelizer.rs:94-106 asserts */
+    if (num_channels <= 2 || !taps || n_taps < num_channels) {
+        set_err("pfb: num_channels > 2 and taps.len() >= num_channels");
+        return nullptr;
+    }
+    if (oversample_rate != 1.0f) {
+        set_err("pfb: only oversample_rate == 1 (maximally decimated) is "
+                "implemented on the GPU path");
+        return nullptr;
+    }
+    if ((num_channels & (num_channels - 1)) != 0 || num_channels < 4 ||
+        num_channels > 4096) {
+        set_err("pfb: num_channels must be a power of two in [4,4096] "
+                "(FFT kernel constraint)");
+        return nullptr;
+    }
+    fsdr_filter* f = create_common(K_MOVAVG /* placeholder, fixed below */);
+    if (!f) return nullptr;
+    f->kind = K_PFB;
+    f->width = num_channels;
+    f->decim = num_channels; /* D == N at oversample 1 */
+    size_t tpf = (n_taps + num_channels - 1) / num_channels;
+    f->history = tpf;
+    f->n_taps = n_taps;
+    /* partition_filter_taps (utilities.rs:5-25): part[i][c] = taps[i+c*N],
+     * zero-padded */
+    std::vector<float> part(num_channels * tpf, 0.f);
+    for (size_t i = 0; i < num_channels; i++) {
+        size_t cnt = 0;
+        for (size_t t = i; t < n_taps; t += num_channels)
+            part[i * tpf + cnt++] = taps[t];
+    }
+    if (hipMalloc(&f->d_taps, part.size() * sizeof(float)) != hipSuccess ||
+        hipMemcpy(f->d_taps, part.data(), part.size() * sizeof(float),
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        set_err("pfb taps upload failed");
+        delete f;
+        return nullptr;
+    }
+    f->sub = fsdr_fft_cf32_create(num_channels, 1, 0, nullptr);
+    if (!f->sub) {
+        delete f;
+        return nullptr;
+    }
+    return f;
+}
+
 extern "C" fsdr_filter* fsdr_moving_avg_create(size_t width,
                                                float decay_factor,
                                                size_t history) {
@@ -1679,6 +1773,7 @@ extern "C" size_t fsdr_filter_item_sizes(const fsdr_filter* f,
 
 extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (!f) return;
+    if (f->sub) fsdr_filter_destroy(f->sub);
     if (f->d_avg) (void)hipFree(f->d_avg);
     if (f->d_taps) (void)hipFree(f->d_taps);
     if (f->d_rtaps) (void)hipFree(f->d_rtaps);
@@ -1970,6 +2065,10 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
             }
             return FSDR_OK;
         }
+        case K_PFB:
+            set_err("pfb channelizer uses fsdr_pfb_channelizer_run_dev "
+                    "(multi-output block)");
+            return FSDR_ERR_INVALID;
         case K_MOVAVG: {
             /* replicate the work() counting loop on the host */
             size_t in_frames = n_in / f->width;
@@ -2027,6 +2126,45 @@ extern "C" int fsdr_filter_host(fsdr_filter* f, const void* in, size_t n_in,
     if (r->produced)
         HIP_TRY(hipMemcpy(out, f->d_out, r->produced * f->item_out,
                           hipMemcpyDeviceToHost));
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_pfb_channelizer_run_dev(fsdr_filter* f,
+                                            const void* d_in, size_t n_in,
+                                            void* d_out,
+                                            size_t out_cap_per_chan,
+                                            void* stream,
+                                            size_t* produced_per_chan) {
+    REQUIRE_GPU();
+    if (!f || f->kind != K_PFB) {
+        set_err("not a pfb channelizer");
+        return FSDR_ERR_INVALID;
+    }
+    hipStream_t st = (hipStream_t)stream;
+    size_t N = f->width, tpf = f->history;
+    size_t prefill = N * tpf;
+    size_t steps = n_in > prefill ? (n_in - prefill) / N : 0;
+    if (steps > out_cap_per_chan) steps = out_cap_per_chan;
+    if (produced_per_chan) *produced_per_chan = steps;
+    if (steps == 0) return FSDR_OK;
+    int rc = ensure_dev(&f->d_in, &f->d_in_bytes, steps * N * 8);
+    if (rc) return rc;
+    int rc2 = ensure_dev(&f->d_out, &f->d_out_bytes, steps * N * 8);
+    if (rc2) return rc2;
+    hipLaunchKernelGGL(k_pfb_dots,
+                       dim3(grid_for((long long)(steps * N), 256)),
+                       dim3(256), 0, st, (const float2*)d_in,
+                       (float2*)f->d_in, (const float*)f->d_taps, (int)N,
+                       (int)tpf, (long long)steps);
+    HIP_TRY(hipGetLastError());
+    rc = launch_fft(f->sub, f->d_in, f->d_out, steps, st);
+    if (rc) return rc;
+    hipLaunchKernelGGL(k_pfb_scatter,
+                       dim3(grid_for((long long)(steps * N), 256)),
+                       dim3(256), 0, st, (const float2*)f->d_out,
+                       (float2*)d_out, (int)N, (long long)steps,
+                       (long long)out_cap_per_chan);
+    HIP_TRY(hipGetLastError());
     return FSDR_OK;
 }
 

@@ -144,6 +144,19 @@ int fsdr_rotator_dev(const void* d_in, void* d_out, size_t n,
                      float phase0_im, void* stream, float* final_re,
                      float* final_im);
 
+/* PfbChannelizer (src/blocks/pfb/channelizer.rs, liquid-dsp scheme):
+ * splits a Complex32 stream into num_channels frequency channels.
+ * GPU path implements the maximally-decimated case (oversample_rate = 1)
+ * in bulk form from zero state; num_channels must be a power of two in
+ * [4,4096]. Output is channel-major: out[c*out_cap_per_chan + k]. */
+fsdr_filter* fsdr_pfb_channelizer_create(size_t num_channels,
+                                         const float* taps, size_t n_taps,
+                                         float oversample_rate);
+int fsdr_pfb_channelizer_run_dev(fsdr_filter* f, const void* d_in,
+                                 size_t n_in, void* d_out,
+                                 size_t out_cap_per_chan, void* stream,
+                                 size_t* produced_per_chan);
+
 /* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
 int fsdr_dev_alloc(void** d_ptr, size_t bytes);
 int fsdr_dev_free(void* d_ptr);

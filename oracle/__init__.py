@@ -93,6 +93,8 @@ def _load():
     lib.oracle_moving_avg.restype = None
     lib.oracle_moving_avg.argtypes = [sz, ctypes.c_float, sz, f32p, szp,
                                       vp, sz, vp, sz, szp, szp]
+    lib.oracle_pfb_channelizer.restype = sz
+    lib.oracle_pfb_channelizer.argtypes = [sz, sz, f32p, sz, vp, sz, vp, sz]
     lib.oracle_chain_cf32.restype = sz
     lib.oracle_chain_cf32.argtypes = [f32p, sz, f32p, sz, sz, sz,
                                       vp, sz, vp, sz, ctypes.c_int]
@@ -294,6 +296,19 @@ def moving_avg(width, decay, history, inp, n_out, avg=None, i_state=0):
                           ctypes.byref(ist), _c(inp), inp.size, _c(out),
                           out.size, ctypes.byref(cons), ctypes.byref(prod))
     return out[:prod.value], cons.value, prod.value, avg, ist.value
+
+
+def pfb_channelizer(num_channels, decim, taps, inp, out_cap_per_chan):
+    """PfbChannelizer one-shot (channelizer.rs restatement). Returns
+    [num_channels, produced] complex array."""
+    lib = _load()
+    taps = np.ascontiguousarray(taps, np.float32)
+    inp = np.ascontiguousarray(inp, CF32)
+    out = np.zeros(num_channels * out_cap_per_chan, CF32)
+    prod = lib.oracle_pfb_channelizer(num_channels, decim, _f32p(taps),
+                                      taps.size, _c(inp), inp.size,
+                                      _c(out), out_cap_per_chan)
+    return out.reshape(num_channels, out_cap_per_chan)[:, :prod]
 
 
 def chain_cf32(taps1, taps2, decim, fft_len, inp, capture=True, nthreads=0):

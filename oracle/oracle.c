@@ -498,6 +498,80 @@ void oracle_moving_avg(size_t width, float decay_factor, size_t history,
     *produced = prod * width;
 }
 
+/* pfb/channelizer.rs one-shot restatement */
+size_t oracle_pfb_channelizer(size_t N, size_t D,
+                              const float* taps, size_t n_taps,
+                              const ocf32* in, size_t n_in,
+                              ocf32* out, size_t out_cap_per_chan) {
+    /* partition_filter_taps (utilities.rs:5-25) */
+    size_t tpf = (n_taps + N - 1) / N;
+    float* part = (float*)calloc(N * tpf, sizeof(float));
+    for (size_t i = 0; i < N; i++) {
+        size_t cnt = 0;
+        for (size_t t = i; t < n_taps; t += N) part[i * tpf + cnt++] = taps[t];
+        /* remaining entries stay 0 (the pad) */
+    }
+    /* window buffers: w[b][0..tpf) oldest->newest (window_buffer.rs) */
+    ocf32* win = (ocf32*)calloc(N * tpf, sizeof(ocf32));
+    size_t* missing = (size_t*)malloc(N * sizeof(size_t));
+    for (size_t b = 0; b < N; b++) missing[b] = tpf;
+    size_t base = N - 1;
+    size_t consumed = 0;
+    /* prefill (:156-180) */
+    int all_filled = 0;
+    while (!all_filled && consumed < n_in) {
+        ocf32* w = win + base * tpf;
+        memmove(w, w + 1, (tpf - 1) * sizeof(ocf32));
+        w[tpf - 1] = in[consumed];
+        if (missing[base]) missing[base]--;
+        base = (base == 0) ? N - 1 : base - 1;
+        consumed++;
+        all_filled = 1;
+        for (size_t b = 0; b < N; b++)
+            if (missing[b]) all_filled = 0;
+    }
+    size_t produced = 0;
+    if (all_filled) {
+        size_t steps = (n_in - consumed) / D;
+        if (steps > out_cap_per_chan) steps = out_cap_per_chan;
+        double* fr = (double*)malloc(N * 2 * sizeof(double));
+        double* fi = fr + N;
+        ocf32* fb = (ocf32*)malloc(N * sizeof(ocf32));
+        for (size_t k = 0; k < steps; k++) {
+            for (size_t j = 0; j < D; j++) { /* :185-191 */
+                ocf32* w = win + base * tpf;
+                memmove(w, w + 1, (tpf - 1) * sizeof(ocf32));
+                w[tpf - 1] = in[consumed + k * D + j];
+                base = (base == 0) ? N - 1 : base - 1;
+            }
+            for (size_t i = 0; i < N; i++) { /* :193-202 */
+                size_t b = (base + i + 1) % N;
+                const ocf32* w = win + b * tpf;
+                float sre = 0.f, sim = 0.f;
+                for (size_t t = 0; t < tpf; t++) {
+                    float tap = part[i * tpf + (tpf - 1 - t)];
+                    sre = sre + w[t].re * tap;
+                    sim = sim + w[t].im * tap;
+                }
+                fb[b].re = sre;
+                fb[b].im = sim;
+            }
+            for (size_t b = 0; b < N; b++) { fr[b] = fb[b].re; fi[b] = fb[b].im; }
+            dft_f64((int)N, 1, fr, fi); /* rustfft inverse, unnormalized */
+            for (size_t cidx = 0; cidx < N; cidx++)
+                out[cidx * out_cap_per_chan + k] =
+                    (ocf32){(float)fr[cidx], (float)fi[cidx]};
+        }
+        produced = steps;
+        free(fr);
+        free(fb);
+    }
+    free(part);
+    free(win);
+    free(missing);
+    return produced;
+}
+
 /* ---------- CPU-baseline chain ---------------------------------------- */
 
 /* f32 iterative radix-2 FFT (forward, unnormalized) for the baseline leg:

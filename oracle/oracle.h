@@ -147,6 +147,20 @@ void oracle_moving_avg(size_t width, float decay_factor, size_t history,
                        float* out, size_t n_out,
                        size_t* consumed, size_t* produced);
 
+/* PfbChannelizer — src/blocks/pfb/channelizer.rs:125-223 (liquid-dsp
+ * scheme): round-robin window buffers (window_buffer.rs), per-channel
+ * sub-filters from partition_filter_taps (utilities.rs:5-25, taps NOT
+ * reversed in partitioning; FirFilter then applies them reversed), one
+ * unnormalized inverse FFT of size num_channels per output step.
+ * One-shot from zero state: prefills, then produces
+ * min(out_cap_per_chan, remaining/decimation) samples per channel.
+ * out is channel-major: out[c*out_cap_per_chan + k]. decimation_factor =
+ * num_channels / oversample_rate (:104-106). */
+size_t oracle_pfb_channelizer(size_t num_channels, size_t decimation_factor,
+                              const float* taps, size_t n_taps,
+                              const ocf32* in, size_t n_in,
+                              ocf32* out, size_t out_cap_per_chan);
+
 /* CPU-baseline chain: FIR(taps1) → decim-by-D(taps2) → per-frame
  * unnormalized fft_len-pt forward DFT (f32 radix-2, same work as the GPU
  * chain), OpenMP-sharded over contiguous chunks with (n_taps-1)-sample

@@ -572,3 +572,28 @@ def test_xlating_fir_parity(gpu, oracle_lib):
     ideal2 = ref * np.exp(1j * theta * (np.arange(p) + 1 + p)).astype(
         np.complex64)
     assert_close(got2, ideal2, 5e-4)
+
+
+# ---------------- PFB channelizer --------------------------------------
+
+@pytest.mark.parametrize("N,n_taps,n_in", [(8, 64, 4096), (4, 37, 2000),
+                                           (16, 128, 8192)])
+def test_pfb_channelizer_parity(gpu, oracle_lib, N, n_taps, n_in):
+    r = rng(N * 1000 + n_taps)
+    taps = r.uniform(-1, 1, n_taps).astype(np.float32)
+    x = cplx(r, n_in)
+    got = gpu.PfbChannelizer(N, taps).run(x)
+    ref = oracle_lib.pfb_channelizer(N, N, taps, x,
+                                     max(1, (n_in) // N))
+    assert got.shape == ref.shape
+    assert_close(got, ref, 1e-4)
+
+
+def test_pfb_channelizer_tone_isolation_gpu(gpu, oracle_lib):
+    N = 8
+    taps = gpu.kaiser_lowpass(1.0 / (2 * N), 0.05, 1e-3)
+    m = np.arange(8192, dtype=np.float64)
+    x = np.exp(2j * np.pi * (3 / N) * m).astype(np.complex64)
+    ch = gpu.PfbChannelizer(N, taps).run(x)
+    e = (np.abs(ch[:, 10:]) ** 2).sum(axis=1)
+    assert e[3] / e.sum() > 0.95
