@@ -1963,7 +1963,9 @@ static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
     int n = (int)f->fft_len;
     int log2n = 0;
     while ((1 << log2n) < n) log2n++;
-    int fpb = 2048 / n; /* 2 frames/block at n=1024: shared barriers */
+    int fpb_base = 1024;
+    if (const char* e = getenv("FSDR_FFT_FPB_BASE")) fpb_base = atoi(e);
+    int fpb = fpb_base / n;
     if (fpb < 1) fpb = 1;
     if (fpb > 16) fpb = 16;
     size_t lds = 2 * (size_t)fpb * n * sizeof(float2);
