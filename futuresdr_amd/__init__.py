@@ -103,6 +103,13 @@ def _load():
     lib.fsdr_filter_dev.restype = ctypes.c_int
     lib.fsdr_filter_dev.argtypes = [vp, vp, sz, vp, sz, vp, rp]
     lib.fsdr_filter_destroy.argtypes = [vp]
+    lib.fsdr_cmul_conj_dev.restype = ctypes.c_int
+    lib.fsdr_cmul_conj_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
+                                       ctypes.POINTER(sz)]
+    lib.fsdr_wlan_moving_sum_dev.restype = ctypes.c_int
+    lib.fsdr_wlan_moving_sum_dev.argtypes = [vp, sz, vp, sz, sz,
+                                             ctypes.c_int, vp,
+                                             ctypes.POINTER(sz)]
     lib.fsdr_cmul_dev.restype = ctypes.c_int
     lib.fsdr_cmul_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
                                   ctypes.POINTER(sz)]
@@ -523,6 +530,64 @@ def lowpass_kaiser_n(n_taps, beta, cutoff):
 
 def kaiser_beta(max_ripple):
     return _load().fsdr_kaiser_beta(max_ripple)
+
+
+def _dev_roundtrip(fn, arrays_in, out_dtype, out_items, *extra):
+    """Helper: upload arrays, call fn(d_ptrs..., d_out, ...), download."""
+    lib = _load()
+    ptrs = []
+    try:
+        for a in arrays_in:
+            p = ctypes.c_void_p()
+            _check(lib.fsdr_dev_alloc(ctypes.byref(p), max(1, a.nbytes)))
+            _check(lib.fsdr_memcpy_h2d(p, ctypes.c_void_p(a.ctypes.data),
+                                       a.nbytes))
+            ptrs.append(p)
+        out = np.zeros(out_items, out_dtype)
+        d_out = ctypes.c_void_p()
+        _check(lib.fsdr_dev_alloc(ctypes.byref(d_out), max(1, out.nbytes)))
+        ptrs.append(d_out)
+        produced = fn(ptrs[:-1], d_out, *extra)
+        _check(lib.fsdr_synchronize())
+        _check(lib.fsdr_memcpy_d2h(ctypes.c_void_p(out.ctypes.data), d_out,
+                                   out.nbytes))
+        return out, produced
+    finally:
+        for p in ptrs:
+            lib.fsdr_dev_free(p)
+
+
+def cmul_conj_host(a, b):
+    lib = _load()
+    a = np.ascontiguousarray(a, CF32)
+    b = np.ascontiguousarray(b, CF32)
+    n = min(a.size, b.size)
+
+    def call(d_ins, d_out):
+        m = ctypes.c_size_t()
+        _check(lib.fsdr_cmul_conj_dev(d_ins[0], a.size, d_ins[1], b.size,
+                                      d_out, n, None, ctypes.byref(m)))
+        return m.value
+
+    out, m = _dev_roundtrip(call, [a, b], CF32, n)
+    return out[:m]
+
+
+def wlan_moving_sum_host(inp, length):
+    lib = _load()
+    is_c = np.iscomplexobj(inp)
+    inp = np.ascontiguousarray(inp, CF32 if is_c else np.float32)
+    n_out = inp.size + length - 1  # pad + sums (fresh state)
+
+    def call(d_ins, d_out):
+        p = ctypes.c_size_t()
+        _check(lib.fsdr_wlan_moving_sum_dev(d_ins[0], inp.size, d_out,
+                                            n_out, length, int(is_c), None,
+                                            ctypes.byref(p)))
+        return p.value
+
+    out, p = _dev_roundtrip(call, [inp], inp.dtype, n_out)
+    return out[:p]
 
 
 def cmul_host(a, b):

@@ -967,6 +967,38 @@ __global__ __launch_bounds__(256) void k_xlating_decim_ccf32(
     }
 }
 
+/* ================= WLAN sync-short helpers ============================ *
+ * examples/wlan/src/bin/rx.rs:73-96 autocorrelation chain pieces:
+ * a*conj(b) Combine (:81) and the sliding-SUM MovingAverage
+ * (moving_average.rs:65-105) with its len-1 zero prologue. */
+__global__ void k_cmul_conj(const float2* __restrict__ a,
+                            const float2* __restrict__ b,
+                            float2* __restrict__ o, long long n) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        float2 x = a[i], y = b[i];
+        o[i] = make_float2(x.x * y.x + x.y * y.y, x.y * y.x - x.x * y.y);
+    }
+}
+
+/* one output per lane: out[i] = sum in[i..i+len) (per float lane);
+ * the zero prologue is emitted by the host wrapper */
+__global__ void k_moving_sum(const float* __restrict__ in,
+                             float* __restrict__ out, int width, int len,
+                             long long n_out_items) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    long long total = n_out_items * width;
+    for (long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         id < total; id += stride) {
+        long long i = id / width;
+        int q = (int)(id - i * width);
+        float sum = 0.f;
+        for (int t = 0; t < len; t++) sum += in[(i + t) * width + q];
+        out[id] = sum;
+    }
+}
+
 /* ================= PFB channelizer (maximally decimated) ============== *
  * src/blocks/pfb/channelizer.rs (liquid-dsp scheme), bulk form for
  * oversample_rate = 1 from zero state: with D == N the round-robin
@@ -2166,6 +2198,48 @@ extern "C" int fsdr_pfb_channelizer_run_dev(fsdr_filter* f,
                        dim3(256), 0, st, (const float2*)f->d_out,
                        (float2*)d_out, (int)N, (long long)steps,
                        (long long)out_cap_per_chan);
+    HIP_TRY(hipGetLastError());
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_cmul_conj_dev(const void* d_a, size_t n_a,
+                                  const void* d_b, size_t n_b, void* d_out,
+                                  size_t n_out, void* stream, size_t* m) {
+    REQUIRE_GPU();
+    size_t mm = n_a < n_b ? n_a : n_b;
+    if (n_out < mm) mm = n_out;
+    if (m) *m = mm;
+    if (mm == 0) return FSDR_OK;
+    hipLaunchKernelGGL(k_cmul_conj, dim3(grid_for((long long)mm, 256)),
+                       dim3(256), 0, (hipStream_t)stream,
+                       (const float2*)d_a, (const float2*)d_b,
+                       (float2*)d_out, (long long)mm);
+    HIP_TRY(hipGetLastError());
+    return FSDR_OK;
+}
+
+/* WLAN MovingAverage, one-shot from fresh state: emits len-1 zero items
+ * then sliding sums; is_complex selects cf32 vs f32 items. */
+extern "C" int fsdr_wlan_moving_sum_dev(const void* d_in, size_t n_in,
+                                        void* d_out, size_t n_out,
+                                        size_t len, int is_complex,
+                                        void* stream, size_t* produced) {
+    REQUIRE_GPU();
+    if (len == 0) { set_err("len must be > 0"); return FSDR_ERR_INVALID; }
+    hipStream_t st = (hipStream_t)stream;
+    size_t w = is_complex ? 2 : 1;
+    size_t pad = len - 1 < n_out ? len - 1 : n_out;
+    if (pad)
+        HIP_TRY(hipMemsetAsync(d_out, 0, pad * w * sizeof(float), st));
+    size_t m = n_in + 1 > len ? n_in + 1 - len : 0;
+    if (m > n_out - pad) m = n_out - pad;
+    if (produced) *produced = pad + m;
+    if (m == 0) return FSDR_OK;
+    hipLaunchKernelGGL(k_moving_sum,
+                       dim3(grid_for((long long)(m * w), 256)), dim3(256),
+                       0, st, (const float*)d_in,
+                       (float*)d_out + pad * w, (int)w, (int)len,
+                       (long long)m);
     HIP_TRY(hipGetLastError());
     return FSDR_OK;
 }

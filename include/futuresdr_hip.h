@@ -157,6 +157,16 @@ int fsdr_pfb_channelizer_run_dev(fsdr_filter* f, const void* d_in,
                                  size_t out_cap_per_chan, void* stream,
                                  size_t* produced_per_chan);
 
+/* WLAN sync-short autocorrelation helpers (examples/wlan/src/bin/
+ * rx.rs:73-96): a*conj(b) Combine and the sliding-SUM MovingAverage
+ * (moving_average.rs:65-105; one-shot: len-1 zero items then sums). */
+int fsdr_cmul_conj_dev(const void* d_a, size_t n_a, const void* d_b,
+                       size_t n_b, void* d_out, size_t n_out, void* stream,
+                       size_t* m);
+int fsdr_wlan_moving_sum_dev(const void* d_in, size_t n_in, void* d_out,
+                             size_t n_out, size_t len, int is_complex,
+                             void* stream, size_t* produced);
+
 /* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
 int fsdr_dev_alloc(void** d_ptr, size_t bytes);
 int fsdr_dev_free(void* d_ptr);

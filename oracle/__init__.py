@@ -95,6 +95,11 @@ def _load():
                                       vp, sz, vp, sz, szp, szp]
     lib.oracle_pfb_channelizer.restype = sz
     lib.oracle_pfb_channelizer.argtypes = [sz, sz, f32p, sz, vp, sz, vp, sz]
+    lib.oracle_wlan_moving_sum.restype = sz
+    lib.oracle_wlan_moving_sum.argtypes = [sz, ctypes.c_int, f32p, sz,
+                                           f32p, sz]
+    lib.oracle_cmul_conj.restype = sz
+    lib.oracle_cmul_conj.argtypes = [vp, sz, vp, sz, vp, sz]
     lib.oracle_chain_cf32.restype = sz
     lib.oracle_chain_cf32.argtypes = [f32p, sz, f32p, sz, sz, sz,
                                       vp, sz, vp, sz, ctypes.c_int]
@@ -296,6 +301,32 @@ def moving_avg(width, decay, history, inp, n_out, avg=None, i_state=0):
                           ctypes.byref(ist), _c(inp), inp.size, _c(out),
                           out.size, ctypes.byref(cons), ctypes.byref(prod))
     return out[:prod.value], cons.value, prod.value, avg, ist.value
+
+
+def wlan_moving_sum(inp, length):
+    """WLAN MovingAverage (sliding sum) one-shot — moving_average.rs."""
+    lib = _load()
+    is_c = np.iscomplexobj(inp)
+    inp = np.ascontiguousarray(inp, CF32 if is_c else np.float32)
+    n_out = inp.size + length - 1
+    out = np.zeros(n_out, inp.dtype)
+    p = lib.oracle_wlan_moving_sum(
+        length, int(is_c),
+        inp.view(np.float32).ctypes.data_as(
+            ctypes.POINTER(ctypes.c_float)), inp.size,
+        out.view(np.float32).ctypes.data_as(
+            ctypes.POINTER(ctypes.c_float)), n_out)
+    return out[:p]
+
+
+def cmul_conj(a, b):
+    lib = _load()
+    a = np.ascontiguousarray(a, CF32)
+    b = np.ascontiguousarray(b, CF32)
+    m = min(a.size, b.size)
+    out = np.zeros(m, CF32)
+    mm = lib.oracle_cmul_conj(_c(a), a.size, _c(b), b.size, _c(out), m)
+    return out[:mm]
 
 
 def pfb_channelizer(num_channels, decim, taps, inp, out_cap_per_chan):

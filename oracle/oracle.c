@@ -572,6 +572,42 @@ size_t oracle_pfb_channelizer(size_t N, size_t D,
     return produced;
 }
 
+size_t oracle_wlan_moving_sum(size_t len, int is_complex,
+                              const float* in, size_t n_in_items,
+                              float* out, size_t n_out_items) {
+    size_t w = is_complex ? 2 : 1;
+    size_t pad = len - 1;
+    size_t prod = 0;
+    while (pad > 0 && prod < n_out_items) { /* :77-84 zero prologue */
+        for (size_t q = 0; q < w; q++) out[prod * w + q] = 0.f;
+        pad--;
+        prod++;
+    }
+    size_t m = n_in_items + 1 > len ? n_in_items + 1 - len : 0;
+    if (m > n_out_items - prod) m = n_out_items - prod;
+    for (size_t q = 0; q < w; q++) { /* :92-99 running sum per lane */
+        float sum = 0.f;
+        for (size_t t = 0; t + 1 < len; t++) sum += in[t * w + q];
+        for (size_t i = 0; i < m; i++) {
+            sum += in[(i + len - 1) * w + q];
+            out[(prod + i) * w + q] = sum;
+            sum -= in[i * w + q];
+        }
+    }
+    return prod + m;
+}
+
+size_t oracle_cmul_conj(const ocf32* a, size_t n_a, const ocf32* b,
+                        size_t n_b, ocf32* out, size_t n_out) {
+    size_t m = n_a < n_b ? n_a : n_b;
+    if (n_out < m) m = n_out;
+    for (size_t i = 0; i < m; i++) {
+        out[i].re = a[i].re * b[i].re + a[i].im * b[i].im;
+        out[i].im = a[i].im * b[i].re - a[i].re * b[i].im;
+    }
+    return m;
+}
+
 /* ---------- CPU-baseline chain ---------------------------------------- */
 
 /* f32 iterative radix-2 FFT (forward, unnormalized) for the baseline leg:

@@ -161,6 +161,18 @@ size_t oracle_pfb_channelizer(size_t num_channels, size_t decimation_factor,
                               const ocf32* in, size_t n_in,
                               ocf32* out, size_t out_cap_per_chan);
 
+/* WLAN MovingAverage (sliding SUM with len-1 zero prologue) —
+ * examples/wlan/src/moving_average.rs:65-105; KAT :117-127 (len 2,
+ * [1,2] -> [0,3]). One-shot from fresh state; item = f32 (width 1) or
+ * cf32 (width 2 floats). Returns produced items. */
+size_t oracle_wlan_moving_sum(size_t len, int is_complex,
+                              const float* in, size_t n_in_items,
+                              float* out, size_t n_out_items);
+
+/* Combine a * conj(b) — examples/wlan/src/bin/rx.rs:81. */
+size_t oracle_cmul_conj(const ocf32* a, size_t n_a, const ocf32* b,
+                        size_t n_b, ocf32* out, size_t n_out);
+
 /* CPU-baseline chain: FIR(taps1) → decim-by-D(taps2) → per-frame
  * unnormalized fft_len-pt forward DFT (f32 radix-2, same work as the GPU
  * chain), OpenMP-sharded over contiguous chunks with (n_taps-1)-sample

@@ -597,3 +597,48 @@ def test_pfb_channelizer_tone_isolation_gpu(gpu, oracle_lib):
     ch = gpu.PfbChannelizer(N, taps).run(x)
     e = (np.abs(ch[:, 10:]) ** 2).sum(axis=1)
     assert e[3] / e.sum() > 0.95
+
+
+# ---------------- WLAN sync-short autocorrelation chain ----------------
+
+def test_wlan_ops_parity(gpu, oracle_lib):
+    r = rng(131)
+    a, b = cplx(r, 3000), cplx(r, 3000)
+    got = gpu.cmul_conj_host(a, b)
+    ref = oracle_lib.cmul_conj(a, b)
+    assert_close(got, ref, 1e-6)
+    x = cplx(r, 5000)
+    got = gpu.wlan_moving_sum_host(x, 48)
+    ref = oracle_lib.wlan_moving_sum(x, 48)
+    assert got.size == ref.size
+    assert_close(got, ref, 1e-5)
+    xf = r.uniform(-1, 1, 5000).astype(np.float32)
+    got = gpu.wlan_moving_sum_host(xf, 64)
+    ref = oracle_lib.wlan_moving_sum(xf, 64)
+    assert_close(got, ref, 1e-5)
+
+
+def test_wlan_sync_short_metric_chain(gpu, oracle_lib):
+    """The rx.rs:73-96 autocorrelation front end on a synthesized 802.11
+    STF-like preamble (16-sample periodic pattern): the correlation
+    metric |movsum48(x*conj(delay16(x)))| / movsum64(|x|^2) plateaus
+    near 1 inside the preamble and stays low in noise. All compute
+    stages on GPU, composed through the ABI."""
+    r = rng(137)
+    stf = cplx(r, 16)  # one short-training symbol
+    preamble = np.tile(stf, 10)  # 160-sample STF
+    noise = 0.1 * cplx(r, 400)
+    sig = np.concatenate([noise[:200], preamble, noise[200:]])
+    n = sig.size
+    # GPU chain
+    delayed = np.concatenate([np.zeros(16, np.complex64), sig[:-16]])
+    prod = gpu.cmul_conj_host(sig, delayed)
+    corr = gpu.wlan_moving_sum_host(prod, 48)[: n]
+    mag = np.abs(sig) ** 2
+    power = gpu.wlan_moving_sum_host(mag.astype(np.float32), 64)[: n]
+    metric = np.abs(corr) / np.maximum(power, 1e-9)
+    # inside the plateau (after warmup of delay+windows)
+    inside = metric[200 + 80:200 + 150]
+    outside = metric[:150]
+    assert inside.min() > 0.6, inside.min()
+    assert np.median(outside) < 0.4, np.median(outside)

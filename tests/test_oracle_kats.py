@@ -297,3 +297,26 @@ def test_pfb_channelizer_linearity(oracle_lib):
     b = o.pfb_channelizer(N, N, taps, x2, 128)
     ab = o.pfb_channelizer(N, N, taps, x1 + x2, 128)
     np.testing.assert_allclose(ab, a + b, atol=1e-4)
+
+
+def test_wlan_moving_sum_kat(oracle_lib):
+    # examples/wlan/src/moving_average.rs:117-127 (mov_avg_one):
+    # len 2, input [1,2] -> output [0, 3]
+    o = oracle_lib
+    out = o.wlan_moving_sum(np.array([1.0, 2.0], np.float32), 2)
+    np.testing.assert_array_equal(out, [0.0, 3.0])
+    # complex sliding sum with prologue
+    x = (np.arange(6) + 1j * np.arange(6)).astype(np.complex64)
+    out = o.wlan_moving_sum(x, 3)
+    ref = np.concatenate([np.zeros(2, np.complex64),
+                          x[0:4] + x[1:5] + x[2:6]])
+    np.testing.assert_allclose(out, ref)
+
+
+def test_cmul_conj_semantics(oracle_lib):
+    o = oracle_lib
+    r = np.random.default_rng(21)
+    a = (r.uniform(-1, 1, (50, 2)) @ [1, 1j]).astype(np.complex64)
+    b = (r.uniform(-1, 1, (40, 2)) @ [1, 1j]).astype(np.complex64)
+    got = o.cmul_conj(a, b)
+    np.testing.assert_allclose(got, a[:40] * np.conj(b), atol=1e-6)
