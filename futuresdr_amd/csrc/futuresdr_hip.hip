@@ -670,12 +670,9 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     const unsigned span = 3 + 4 * elemsP; /* input elements per tile */
     constexpr int NL2 =
         (2 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
-    float2 stgA[NL2];
+    float2 stgA[NL2], stgB[NL2];
     /* elements of phases {2h, 2h+1}: rel = 3 + 4i + 2h + vloc */
     auto load_half = [&](long long tl, int h, float2 (&stg)[NL2]) {
-        /* single staging array: each load_half's results are consumed by
-         * the next write_half after a barrier; the compiler's counted
-         * vmcnt keeps the loads in flight under the interleaved MFMAs */
         const long long ib = tl * MDFIR_TILE * 4;
 #pragma unroll
         for (int j = 0; j < NL2; j++) {
@@ -729,10 +726,10 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
         v4f cim = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
-        load_half(tile, 1, stgA);     /* in flight under half-0 MFMAs */
+        load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
         mfma_half(0, cre, cim);
         __syncthreads();
-        write_half(stgA);
+        write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
