@@ -103,6 +103,15 @@ __device__ __host__ __forceinline__ unsigned plane_floats(unsigned elems) {
     return (lds_pad(elems - 1) + 4u) & ~3u;
 }
 
+/* LDS index swizzle for the ping/pong buffers: float2 element i sits at
+ * bank (2i) mod 64, so accesses whose lane stride is a multiple of 32
+ * elements (the strided Stockham stages) are up-to-8-way conflicted;
+ * XORing bits 5..7 of the element index into bits 2..4 makes them
+ * conflict-free while keeping contiguous stages conflict-free. */
+__device__ __forceinline__ unsigned fft_swz(unsigned i) {
+    return i ^ (((i >> 5) & 7u) << 2);
+}
+
 __device__ __forceinline__ float2 f2_add(float2 a, float2 b) {
     return make_float2(a.x + b.x, a.y + b.y);
 }
@@ -639,6 +648,12 @@ __global__ __launch_bounds__(MFIR32_BLOCK) void k_fir_mfma32_tpl(
 #define MDFIR_BLOCK 256
 #define MDFIR_TILE 1024 /* decimated outputs per block */
 
+/* forward declaration (defined with the FFT kernel below): in-block
+ * 1024-pt forward Stockham FFT over swizzled LDS ping/pong buffers */
+__device__ void fft1024_block(float2* ping, float2* pong,
+                              const float2* __restrict__ twid);
+/* result lands in `pong` (5 stages = 5 swaps); read pong[fft_swz(i)] */
+
 template <int KKD>
 __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     const float2* __restrict__ in, float2* __restrict__ out,
@@ -739,6 +754,127 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
             int row = k4 * 4 + q;
             long long o = out_base + (long long)wave * 256 + 16 * row + r16;
             if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+    }
+}
+
+template <int KKD>
+__global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv /* [4][KKD], rtv[v][u] = rt[4u+v] */,
+    long long n_out, long long n_in_valid,
+    const float2* __restrict__ twid /* 1024-entry forward table */,
+    float* __restrict__ mag_out /* nullable |X|^2 */) {
+    static_assert(MDFIR_TILE == 1024, "one tile == one 1024-pt FFT frame");
+    static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
+    const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    /* two phases resident at a time ([re_v0, re_v1, im_v0, im_v1]) —
+     * halves LDS vs all-phase planes, doubling resident blocks/CU;
+     * accumulators carry across the two halves, and each half's global
+     * loads are issued under the other half's MFMAs. */
+    float* planes = (float*)smem;        /* [4][SPm] */
+    float* s_rtx = planes + 4u * SPm;    /* [4][KKD+16], 15-zero prologue */
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+
+    for (int i = tid; i < 4 * (KKD + 16); i += MDFIR_BLOCK) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    __syncthreads();
+
+    const unsigned span = 3 + 4 * elemsP; /* input elements per tile */
+    constexpr int NL2 =
+        (2 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float2 stgA[NL2], stgB[NL2];
+    /* elements of phases {2h, 2h+1}: rel = 3 + 4i + 2h + vloc */
+    auto load_half = [&](long long tl, int h, float2 (&stg)[NL2]) {
+        const long long ib = tl * MDFIR_TILE * 4;
+#pragma unroll
+        for (int j = 0; j < NL2; j++) {
+            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+            unsigned rel = 3 + 4 * (idx >> 1) + 2 * h + (idx & 1u);
+            long long g = ib + rel;
+            stg[j] = (rel < span && g < n_in_valid)
+                         ? in[g] : make_float2(0.f, 0.f);
+        }
+    };
+    auto write_half = [&](const float2 (&stg)[NL2]) {
+#pragma unroll
+        for (int j = 0; j < NL2; j++) {
+            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+            unsigned i = idx >> 1, vloc = idx & 1u;
+            if (i < elemsP) {
+                unsigned d = mfma_swz(i);
+                planes[vloc * SPm + d] = stg[j].x;
+                planes[(2 + vloc) * SPm + d] = stg[j].y;
+            }
+        }
+    };
+    const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
+#pragma unroll
+        for (int vloc = 0; vloc < 2; vloc++) {
+            const float* pre = planes + (unsigned)vloc * SPm;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm;
+            const int v = 2 * h + vloc;
+            float bfrag[KKD / 4];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++)
+                bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++) {
+                float a_re = pre[mfma_swz(ab + 4 * s)];
+                float a_im = pim[mfma_swz(ab + 4 * s)];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
+            }
+        }
+    };
+
+    load_half(blockIdx.x, 0, stgA);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MDFIR_TILE;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        write_half(stgA);
+        __syncthreads();
+        load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
+        mfma_half(0, cre, cim);
+        __syncthreads();
+        write_half(stgB);
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
+        mfma_half(1, cre, cim);
+        __syncthreads(); /* phase planes are dead; reuse them as FFT LDS */
+        float2* ping = (float2*)planes;       /* 1024 float2 = 8 KB */
+        float2* pong = ping + 1024;           /* fits in 4*SPm floats */
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            int pos = wave * 256 + 16 * row + r16; /* y2 index in tile */
+            ping[fft_swz((unsigned)pos)] = make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+        fft1024_block(ping, pong, twid); /* result in pong (swizzled) */
+        for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+            long long o = out_base + i;
+            if (o < n_out) {
+                float2 v = pong[fft_swz((unsigned)i)];
+                out[o] = v;
+                if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+            }
         }
         __syncthreads();
     }
@@ -1171,13 +1307,45 @@ __device__ __forceinline__ float2 cmul_tw(float2 a, float2 w, int inverse) {
     return cmulf(a, w);
 }
 
-/* LDS index swizzle for the ping/pong buffers: float2 element i sits at
- * bank (2i) mod 64, so accesses whose lane stride is a multiple of 32
- * elements (the strided Stockham stages) are up-to-8-way conflicted;
- * XORing bits 5..7 of the element index into bits 2..4 makes them
- * conflict-free while keeping contiguous stages conflict-free. */
-__device__ __forceinline__ unsigned fft_swz(unsigned i) {
-    return i ^ (((i >> 5) & 7u) << 2);
+/* In-block 1024-pt forward FFT (unnormalized), 256 threads, for the
+ * chain's fused decim+FFT kernel. Same radix-4 DIF scheme and fft_swz
+ * LDS swizzle as k_fft_stockham. */
+__device__ void fft1024_block(float2* ping, float2* pong,
+                              const float2* __restrict__ twid) {
+    const int n = 1024;
+    const int tf = threadIdx.x;
+    float2* a = ping;
+    float2* b = pong;
+    int scur = 1;
+    int ncur = n;
+    while (ncur >= 4) {
+        const int m4 = ncur >> 2;
+        for (int bf = tf; bf < n / 4; bf += 256) {
+            const int p = bf / scur;
+            const int q = bf - p * scur;
+            const int tw = n / ncur;
+            float2 x0 = a[fft_swz(q + scur * p)];
+            float2 x1 = a[fft_swz(q + scur * (p + m4))];
+            float2 x2 = a[fft_swz(q + scur * (p + 2 * m4))];
+            float2 x3 = a[fft_swz(q + scur * (p + 3 * m4))];
+            float2 e0 = f2_add(x0, x2), e1 = f2_sub(x0, x2);
+            float2 o0 = f2_add(x1, x3), o1 = f2_sub(x1, x3);
+            float2 o1r = make_float2(o1.y, -o1.x);
+            b[fft_swz(q + scur * (4 * p + 0))] = f2_add(e0, o0);
+            b[fft_swz(q + scur * (4 * p + 1))] =
+                cmulf(f2_add(e1, o1r), twid[(size_t)p * tw]);
+            b[fft_swz(q + scur * (4 * p + 2))] =
+                cmulf(f2_sub(e0, o0), twid[(size_t)2 * p * tw]);
+            b[fft_swz(q + scur * (4 * p + 3))] =
+                cmulf(f2_sub(e1, o1r), twid[(size_t)3 * p * tw]);
+        }
+        float2* t = a; a = b; b = t;
+        scur <<= 2;
+        ncur >>= 2;
+        __syncthreads();
+    }
+    /* 5 radix-4 stages = 5 ping/pong swaps: the result is in `pong`
+     * (callers read pong[fft_swz(i)]). */
 }
 
 __global__ __launch_bounds__(256) void k_fft_stockham(
@@ -2496,6 +2664,27 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                         (prod + 8) * sizeof(float2));
         if (rc) return rc;
         out2 = c->d_null;
+    }
+    const char* ffz = getenv("FSDR_CHAIN_FFTFUSE");
+    if (c->fused && L == 1024 && c->fused->kk_mfma == 80 &&
+        !c->fft->inverse && !c->fft->fft_shift && c->fft->norm == 0.f &&
+        (!ffz || atoi(ffz) != 0)) {
+        /* single kernel: fused 253-tap decimating filter + in-block
+         * 1024-pt FFT (+ optional |X|^2) — y2 never touches HBM */
+        long long tiles = ((long long)prod + MDFIR_TILE - 1) / MDFIR_TILE;
+        long long cap = 256 * 64;
+        if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
+        int grid = (int)std::min<long long>(tiles, cap);
+        unsigned elemsP = MDFIR_TILE + 80 + 8;
+        size_t lds = (4 * (size_t)((elemsP + 31u) & ~31u) + 4 * (80 + 16)) *
+                     sizeof(float);
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_tpl<80>),
+                           dim3(grid), dim3(MDFIR_BLOCK), lds, st,
+                           (const float2*)d_in, out2, c->fused->d_mtaps,
+                           (long long)prod, (long long)n_in,
+                           (const float2*)c->fft->d_twid, (float*)d_mag);
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
     }
     if (c->fused) {
         rc = launch_decim_cf32(c->fused, d_in, c->d_y2, prod, n_in, st);
