@@ -61,54 +61,52 @@ def alloc_dev(lib, bytes_):
     return p
 
 
-def measure_chain_roofline(fa, torch, d_in, n_samples, taps1, taps2,
-                           decim, traffic_file):
-    """Dominant-kernel roofline: the chain's dominant kernel is the FUSED
-    combined-taps decimating FIR (k_decim4_mfma_tpl<80>).
-    Duration via HIP events on the launch stream (torch's current
-    stream); achieved = the kernel's algorithmic flops per launch
-    ((T1+T2-1) taps x 4 flops per decimated output) / duration."""
-    lib = fa.lib()
+def measure_chain_roofline(fa, torch, chain, d_in, n_samples, taps1,
+                           taps2, decim, fft_len, traffic_file):
+    """Dominant-kernel roofline. The default chain is ONE kernel
+    (k_decim4_fft_mfma_tpl<80>: the algebraically-fused 253-tap
+    decimating filter with an in-block 1024-pt FFT — DESIGN.md §d), so
+    the launch IS the dominant kernel; duration via HIP events on the
+    launch stream. achieved = algorithmic flops per launch: fused-filter
+    (T1+T2-1 taps x 4 flops per decimated output) + FFT (5 N log2 N per
+    frame), all fp32."""
     st = torch.cuda.current_stream()
-    g = np.convolve(taps1.astype(np.float64),
-                    taps2.astype(np.float64)).astype(np.float32)
-    fused = fa.DecimFir(decim, g)
-    produced = (n_samples + 1 - g.size) // decim
-    d_y2 = alloc_dev(lib, (produced + 8) * 8)
-    try:
-        for _ in range(3):
-            fused.filter_dev(d_in.value, n_samples, d_y2.value, produced,
-                             stream=st.cuda_stream)
-        torch.cuda.synchronize()
-        reps = 20
-        ev0 = torch.cuda.Event(enable_timing=True)
-        ev1 = torch.cuda.Event(enable_timing=True)
-        ev0.record(st)
-        for _ in range(reps):
-            fused.filter_dev(d_in.value, n_samples, d_y2.value, produced,
-                             stream=st.cuda_stream)
-        ev1.record(st)
-        torch.cuda.synchronize()
-        ms = ev0.elapsed_time(ev1) / reps
-    finally:
-        lib.fsdr_dev_free(d_y2)
-    flops = produced * g.size * 4
+    g_len = taps1.size + taps2.size - 1
+    produced = (n_samples + 1 - g_len) // decim
+    frames = produced // fft_len
+    prod = frames * fft_len
+    for _ in range(3):
+        chain.run_dev(d_in.value, n_samples, 0, 0, 0, 0,
+                      stream=st.cuda_stream)
+    torch.cuda.synchronize()
+    reps = 20
+    ev0 = torch.cuda.Event(enable_timing=True)
+    ev1 = torch.cuda.Event(enable_timing=True)
+    ev0.record(st)
+    for _ in range(reps):
+        chain.run_dev(d_in.value, n_samples, 0, 0, 0, 0,
+                      stream=st.cuda_stream)
+    ev1.record(st)
+    torch.cuda.synchronize()
+    ms = ev0.elapsed_time(ev1) / reps
+    import math
+    flops = prod * g_len * 4 + frames * 5 * fft_len * math.log2(fft_len)
     achieved_tf = flops / (ms * 1e-3) / 1e12
     traffic = None
     if traffic_file and os.path.exists(traffic_file):
         with open(traffic_file) as f:
             t = json.load(f)
-        bps = t.get("fused_decim_hbm_bytes_per_input_sample")
+        bps = t.get("chain_kernel_hbm_bytes_per_input_sample")
         if bps is not None:
-            traffic = bps * produced * decim
+            traffic = bps * n_samples
     return {
-        "bound": "mfma",  # fp32 MFMA compute-bound: 1012 flops vs ~34 B/out
+        "bound": "mfma",  # fp32 MFMA compute-bound (see DESIGN.md §d)
         "achieved": round(achieved_tf, 2),
         "peak": FP32_PEAK_TFLOPS,
         "unit": "TFLOP/s",
         "frac": round(achieved_tf / FP32_PEAK_TFLOPS, 4),
         "traffic": traffic,
-        "kernel": "k_decim4_mfma_tpl<80> (fused fir127+decim4-127)",
+        "kernel": "k_decim4_fft_mfma_tpl<80> (fused fir+decim+fft+mag)",
         "ms_per_launch": round(ms, 4),
     }
 
@@ -217,9 +215,9 @@ def main():
     value = n_gpus * S * args.steps / elapsed / 1e6  # whole-job MSample/s
 
     if rank == 0:
-        roofline = measure_chain_roofline(fa, torch, d_in, S, taps1,
-                                          taps2, args.decim,
-                                          args.traffic_file)
+        roofline = measure_chain_roofline(fa, torch, chain, d_in, S,
+                                          taps1, taps2, args.decim,
+                                          args.fft, args.traffic_file)
         cpu_baseline = None
         if n_gpus == 1 and not args.skip_cpu_baseline:
             log("measuring CPU baseline (oracle chain, all cores)...")
