@@ -2672,7 +2672,7 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
         /* single kernel: fused 253-tap decimating filter + in-block
          * 1024-pt FFT (+ optional |X|^2) — y2 never touches HBM */
         long long tiles = ((long long)prod + MDFIR_TILE - 1) / MDFIR_TILE;
-        long long cap = 256 * 64;
+        long long cap = 8192; /* ~2 tiles/block at 2^26: best measured */
         if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
         int grid = (int)std::min<long long>(tiles, cap);
         unsigned elemsP = MDFIR_TILE + 80 + 8;
