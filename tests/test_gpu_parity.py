@@ -642,3 +642,41 @@ def test_wlan_sync_short_metric_chain(gpu, oracle_lib):
     outside = metric[:150]
     assert inside.min() > 0.6, inside.min()
     assert np.median(outside) < 0.4, np.median(outside)
+
+
+# ---------------- randomized chain fuzz --------------------------------
+
+@pytest.mark.parametrize("seed", range(6))
+def test_chain_fuzz_vs_oracle(gpu, oracle_lib, seed):
+    """Randomized tap counts / FFT lengths / stream sizes through the
+    fused chain vs the two-stage oracle (covers both the single-kernel
+    fft_len==1024 path and the split path)."""
+    r = rng(10_000 + seed)
+    nt1 = int(r.integers(8, 250))
+    nt2 = int(r.integers(8, 250))
+    fft_len = int(r.choice([256, 512, 1024]))
+    n_in = int(r.integers(4 * fft_len * 2 + nt1 + nt2, 200_000))
+    t1 = r.uniform(-1, 1, nt1).astype(np.float32)
+    t2 = r.uniform(-1, 1, nt2).astype(np.float32)
+    x = cplx(r, n_in)
+    lib = gpu.lib()
+    d_in = ctypes.c_void_p()
+    d_out = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_in), n_in * 8) == 0
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), n_in * 8) == 0
+    try:
+        lib.fsdr_memcpy_h2d(d_in, ctypes.c_void_p(x.ctypes.data), n_in * 8)
+        ch = gpu.Chain(t1, t2, 4, fft_len)
+        cons, prod = ch.run_dev(d_in.value, n_in, d_out.value, n_in)
+        gpu.synchronize()
+        ref, cons_ref = oracle_lib.chain_cf32(t1, t2, 4, fft_len, x)
+        assert (cons, prod) == (cons_ref, ref.size), (nt1, nt2, fft_len)
+        got = np.zeros(prod, np.complex64)
+        lib.fsdr_memcpy_d2h(ctypes.c_void_p(got.ctypes.data), d_out,
+                            prod * 8)
+        if prod:
+            rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+            assert rel < 2e-4, (nt1, nt2, fft_len, rel)
+    finally:
+        lib.fsdr_dev_free(d_in)
+        lib.fsdr_dev_free(d_out)
