@@ -106,6 +106,9 @@ def _load():
     lib.fsdr_cmul_conj_dev.restype = ctypes.c_int
     lib.fsdr_cmul_conj_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
                                        ctypes.POINTER(sz)]
+    lib.fsdr_divide_mag_dev.restype = ctypes.c_int
+    lib.fsdr_divide_mag_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
+                                        ctypes.POINTER(sz)]
     lib.fsdr_wlan_moving_sum_dev.restype = ctypes.c_int
     lib.fsdr_wlan_moving_sum_dev.argtypes = [vp, sz, vp, sz, sz,
                                              ctypes.c_int, vp,
@@ -555,6 +558,22 @@ def _dev_roundtrip(fn, arrays_in, out_dtype, out_items, *extra):
     finally:
         for p in ptrs:
             lib.fsdr_dev_free(p)
+
+
+def divide_mag_host(a, b):
+    lib = _load()
+    a = np.ascontiguousarray(a, CF32)
+    b = np.ascontiguousarray(b, np.float32)
+    n = min(a.size, b.size)
+
+    def call(d_ins, d_out):
+        m = ctypes.c_size_t()
+        _check(lib.fsdr_divide_mag_dev(d_ins[0], a.size, d_ins[1], b.size,
+                                       d_out, n, None, ctypes.byref(m)))
+        return m.value
+
+    out, m = _dev_roundtrip(call, [a, b], np.float32, n)
+    return out[:m]
 
 
 def cmul_conj_host(a, b):

@@ -1118,6 +1118,18 @@ __global__ void k_cmul_conj(const float2* __restrict__ a,
     }
 }
 
+/* divide_mag Combine — rx.rs:97: out = |a| / b (cf32, f32) -> f32 */
+__global__ void k_divide_mag(const float2* __restrict__ a,
+                             const float* __restrict__ b,
+                             float* __restrict__ o, long long n) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        float2 x = a[i];
+        o[i] = sqrtf(x.x * x.x + x.y * x.y) / b[i];
+    }
+}
+
 /* one output per lane: out[i] = sum in[i..i+len) (per float lane);
  * the zero prologue is emitted by the host wrapper */
 __global__ void k_moving_sum(const float* __restrict__ in,
@@ -2366,6 +2378,23 @@ extern "C" int fsdr_pfb_channelizer_run_dev(fsdr_filter* f,
                        dim3(256), 0, st, (const float2*)f->d_out,
                        (float2*)d_out, (int)N, (long long)steps,
                        (long long)out_cap_per_chan);
+    HIP_TRY(hipGetLastError());
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_divide_mag_dev(const void* d_a, size_t n_a,
+                                   const void* d_b, size_t n_b,
+                                   void* d_out, size_t n_out, void* stream,
+                                   size_t* m) {
+    REQUIRE_GPU();
+    size_t mm = n_a < n_b ? n_a : n_b;
+    if (n_out < mm) mm = n_out;
+    if (m) *m = mm;
+    if (mm == 0) return FSDR_OK;
+    hipLaunchKernelGGL(k_divide_mag, dim3(grid_for((long long)mm, 256)),
+                       dim3(256), 0, (hipStream_t)stream,
+                       (const float2*)d_a, (const float*)d_b,
+                       (float*)d_out, (long long)mm);
     HIP_TRY(hipGetLastError());
     return FSDR_OK;
 }

@@ -163,6 +163,10 @@ int fsdr_pfb_channelizer_run_dev(fsdr_filter* f, const void* d_in,
 int fsdr_cmul_conj_dev(const void* d_a, size_t n_a, const void* d_b,
                        size_t n_b, void* d_out, size_t n_out, void* stream,
                        size_t* m);
+/* divide_mag Combine (rx.rs:97): out = |a| / b. */
+int fsdr_divide_mag_dev(const void* d_a, size_t n_a, const void* d_b,
+                        size_t n_b, void* d_out, size_t n_out, void* stream,
+                        size_t* m);
 int fsdr_wlan_moving_sum_dev(const void* d_in, size_t n_in, void* d_out,
                              size_t n_out, size_t len, int is_complex,
                              void* stream, size_t* produced);

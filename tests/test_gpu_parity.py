@@ -680,3 +680,12 @@ def test_chain_fuzz_vs_oracle(gpu, oracle_lib, seed):
     finally:
         lib.fsdr_dev_free(d_in)
         lib.fsdr_dev_free(d_out)
+
+
+def test_divide_mag_parity(gpu):
+    r = rng(139)
+    a = cplx(r, 4000)
+    b = (r.uniform(0.1, 2.0, 4000)).astype(np.float32)
+    got = gpu.divide_mag_host(a, b)
+    ref = (np.abs(a) / b).astype(np.float32)
+    assert_close(got, ref, 1e-5)
