@@ -689,3 +689,60 @@ def test_divide_mag_parity(gpu):
     got = gpu.divide_mag_host(a, b)
     ref = (np.abs(a) / b).astype(np.float32)
     assert_close(got, ref, 1e-5)
+
+
+def test_ring_chain_streaming(gpu, oracle_lib):
+    """The production streaming path: host chunks flow through the
+    pinned-host ring (Slab semantics, 252-sample history prefix) into the
+    fused chain kernel; concatenated spectra must equal the one-shot
+    oracle chain on the zero-history-prefixed stream. Chunk size is a
+    multiple of 4*fft_len so every chunk's decimated output is whole
+    frames (the chain consumes each chunk fully, like the reference block
+    leaves no aligned leftovers)."""
+    lib = gpu.lib()
+    r = rng(149)
+    t1 = r.uniform(-1, 1, 127).astype(np.float32)
+    t2 = r.uniform(-1, 1, 127).astype(np.float32)
+    fft_len = 256
+    chunk = 4 * fft_len * 16            # 16384 samples/chunk
+    n_total = chunk * 8
+    x = cplx(r, n_total)
+    g_len = t1.size + t2.size - 1       # fused taps
+    reserved = g_len - 1                # 252
+    ring = lib.fsdr_ring_create(4, chunk, 8, reserved)
+    assert ring
+    chain = gpu.Chain(t1, t2, 4, fft_len)
+    d_out = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), chunk * 8) == 0
+    outs = []
+    try:
+        for off in range(0, n_total, chunk):
+            hp = ctypes.c_void_p()
+            items = ctypes.c_size_t()
+            assert lib.fsdr_ring_writer_acquire(
+                ring, ctypes.byref(hp), ctypes.byref(items)) == 0
+            ctypes.memmove(hp, ctypes.c_void_p(
+                x[off:off + chunk].ctypes.data), chunk * 8)
+            assert lib.fsdr_ring_writer_commit(ring, chunk) == 0
+            dp = ctypes.c_void_p()
+            got_items = ctypes.c_size_t()
+            assert lib.fsdr_ring_reader_acquire(
+                ring, ctypes.byref(dp), ctypes.byref(got_items)) == 0
+            n_in = reserved + got_items.value
+            cons, prod = chain.run_dev(dp.value, n_in, d_out.value, chunk)
+            gpu.synchronize()
+            assert cons == got_items.value  # whole chunk consumed
+            h = np.zeros(prod, np.complex64)
+            lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
+                                prod * 8)
+            outs.append(h)
+            assert lib.fsdr_ring_reader_release(ring) == 0
+        got = np.concatenate(outs)
+        xz = np.concatenate([np.zeros(reserved, np.complex64), x])
+        ref, _ = oracle_lib.chain_cf32(t1, t2, 4, fft_len, xz)
+        assert got.size == ref.size
+        rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+        assert rel < 2e-4, rel
+    finally:
+        lib.fsdr_dev_free(d_out)
+        lib.fsdr_ring_destroy(ring)
