@@ -721,6 +721,9 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
 #pragma unroll
             for (int s = 0; s < KKD / 4; s++)
                 bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+            __builtin_amdgcn_s_setprio(1); /* boost MFMA waves over the
+                                              FFT/staging phases of
+                                              co-resident blocks (T5) */
 #pragma unroll
             for (int s = 0; s < KKD / 4; s++) {
                 float a_re = pre[mfma_swz(ab + 4 * s)];
@@ -730,6 +733,7 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
                 cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
                     a_im, bfrag[s], cim, 0, 0, 0);
             }
+            __builtin_amdgcn_s_setprio(0);
         }
     };
 
@@ -829,6 +833,9 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
 #pragma unroll
             for (int s = 0; s < KKD / 4; s++)
                 bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+            __builtin_amdgcn_s_setprio(1); /* boost MFMA waves over the
+                                              FFT/staging phases of
+                                              co-resident blocks (T5) */
 #pragma unroll
             for (int s = 0; s < KKD / 4; s++) {
                 float a_re = pre[mfma_swz(ab + 4 * s)];
@@ -838,6 +845,7 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
                 cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
                     a_im, bfrag[s], cim, 0, 0, 0);
             }
+            __builtin_amdgcn_s_setprio(0);
         }
     };
 
