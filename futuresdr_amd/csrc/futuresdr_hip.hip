@@ -1579,7 +1579,8 @@ struct fsdr_filter {
     size_t width = 0;        /* MovingAvg */
     size_t history = 0;
     size_t i_state = 0;
-    float decay = 0.f;
+    float decay = 0.f;        /* MovingAvg decay factor */
+    float theta = 0.f;        /* XlatingFir rotator increment */
     float rot_re = 1.f, rot_im = 0.f; /* xlating rotator phase state */
     fsdr_filter* sub = nullptr;       /* channelizer's internal IFFT */
     float* d_avg = nullptr;
@@ -1895,8 +1896,8 @@ extern "C" fsdr_filter* fsdr_xlating_fir_cf32_create(const float* taps,
         return nullptr;
     }
     /* rotator increment (:91-94) and unit start phase (rotator.rs:17-19) */
-    f->decay = -6.2831853071795864769f * offset * (float)decimation /
-               sample_rate;                     /* reuse field as theta */
+    f->theta = -6.2831853071795864769f * offset * (float)decimation /
+               sample_rate;
     f->rot_re = 1.0f;
     f->rot_im = 0.0f;
     f->n_taps_padded = (int)n_taps;
@@ -1923,9 +1924,8 @@ extern "C" fsdr_filter* fsdr_pfb_channelizer_create(size_t num_channels,
                 "(FFT kernel constraint)");
         return nullptr;
     }
-    fsdr_filter* f = create_common(K_MOVAVG /* placeholder, fixed below */);
+    fsdr_filter* f = create_common(K_PFB);
     if (!f) return nullptr;
-    f->kind = K_PFB;
     f->width = num_channels;
     f->decim = num_channels; /* D == N at oversample 1 */
     size_t tpf = (n_taps + num_channels - 1) / num_channels;
