@@ -2275,10 +2275,10 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                                (float2*)d_out, (const float2*)f->d_taps,
                                (int)f->n_taps, (long long)f->decim,
                                (long long)r->produced, (long long)n_in,
-                               f->decay, f->rot_re, f->rot_im);
+                               f->theta, f->rot_re, f->rot_im);
             HIP_TRY(hipGetLastError());
             { /* advance the rotator phase (closed form, f64) */
-                double a = (double)f->decay * (double)r->produced;
+                double a = (double)f->theta * (double)r->produced;
                 double cs = cos(a), sn = sin(a);
                 float nr = (float)(cs * f->rot_re - sn * f->rot_im);
                 float ni = (float)(cs * f->rot_im + sn * f->rot_re);
