@@ -4,9 +4,14 @@ the 127-tap FIR -> decim-4 -> 1024-pt FFT chain, on N GPUs of one node.
 
 A step = one pass of the chain over one HBM-resident batch of synthetic
 samples (re/im iid uniform[-1,1), device-generated, seeded). At N>1 each
-rank runs an independent channel on its own GPU (config 4) and the 1024-bin
-magnitude spectra are joined with one RCCL all-gather per step (batched,
->= 4 MiB per rank). Weak scaling. Launch for N>1:
+rank runs an independent channel on its own GPU (config 4) and the
+spectrum-combine step joins the ranks' spectra with one RCCL all-gather
+per step. The payload is the reference spectrum SINK's output — the
+MovingAvg-averaged 1024-bin magnitude spectrum (moving_avg.rs is what the
+spectrum app displays), one averaged frame per rank per step (4 KiB) —
+computed on-GPU from every frame's |X|^2. Gathering every RAW frame
+instead (= benchmarking the interconnect, ~67 MB/rank/step) is available
+with --gather-frames. Weak scaling. Launch for N>1:
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
 
@@ -46,6 +51,9 @@ def parse_args():
     p.add_argument("--fft", type=int, default=1024)
     p.add_argument("--decim", type=int, default=4)
     p.add_argument("--taps", type=int, default=127)
+    p.add_argument("--gather-frames", action="store_true",
+                   help="all-gather every raw magnitude frame instead of "
+                        "the MovingAvg-averaged spectrum")
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--cpu-sample", type=int, default=0,
                    help="fixed CPU-baseline sample size (0 = auto ~10s)")
@@ -181,16 +189,25 @@ def main():
     fa.fill_uniform_dev(d_in.value, S, seed=1000 + rank,
                         stream=st.cuda_stream)
     mag = torch.empty(prod, dtype=torch.float32, device="cuda")
+    # spectrum sink: per-bin EMA over the step's frames, emitting one
+    # averaged spectrum per step (MovingAvg(width=fft, decay, history=
+    # frames) — the reference sink's output cadence for this batch size)
+    avg_sink = fa.MovingAvg(args.fft, 0.1, frames)
+    avg_spec = torch.empty(args.fft, dtype=torch.float32, device="cuda")
     gathered = None
     if world > 1:
-        gathered = torch.empty(world * prod, dtype=torch.float32,
+        payload = prod if args.gather_frames else args.fft
+        gathered = torch.empty(world * payload, dtype=torch.float32,
                                device="cuda")
 
     def step():
         chain.run_dev(d_in.value, S, 0, 0, mag.data_ptr(), prod,
                       stream=st.cuda_stream)
+        avg_sink.filter_dev(mag.data_ptr(), prod, avg_spec.data_ptr(),
+                            args.fft, stream=st.cuda_stream)
         if td is not None:
-            td.all_gather_into_tensor(gathered, mag)
+            td.all_gather_into_tensor(
+                gathered, mag if args.gather_frames else avg_spec)
 
     for _ in range(args.warmup):
         step()
@@ -246,7 +263,10 @@ def main():
                 "samples_per_step_per_gpu": S,
                 "frames_per_step_per_gpu": frames,
                 "parallelism": f"{n_gpus} independent channels"
-                               + (" + RCCL all-gather of |X|^2"
+                               + ((" + RCCL all-gather of raw |X|^2 frames"
+                                   if args.gather_frames else
+                                   " + RCCL all-gather of the averaged "
+                                   "spectrum (the MovingAvg sink output)")
                                   if n_gpus > 1 else ""),
             },
             "roofline": roofline,
