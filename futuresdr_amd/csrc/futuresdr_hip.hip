@@ -1073,6 +1073,56 @@ __global__ __launch_bounds__(256) void k_fir_ccf32(
     }
 }
 
+/* Parallel MovingAvg fast path (used when at most one emission occurs,
+ * at the end of the processed frames — e.g. the bench's spectrum sink
+ * with history == frames): the EMA recurrence is chunked over frames and
+ * the per-chunk EMAs composed exactly:
+ *   ema(chunk_0..c) = (1-d)^len_c * ema(chunk_0..c-1) + partial_c. */
+__global__ void k_moving_avg_chunks(const float* __restrict__ in,
+                                    float* __restrict__ partial, int width,
+                                    long long frames, int n_chunks,
+                                    int chunk_frames, float decay) {
+    long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    long long total = (long long)n_chunks * width;
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (; id < total; id += stride) {
+        int c = (int)(id / width);
+        int b = (int)(id - (long long)c * width);
+        long long f0 = (long long)c * chunk_frames;
+        long long f1 = f0 + chunk_frames;
+        if (f1 > frames) f1 = frames;
+        float p = 0.f;
+        for (long long f = f0; f < f1; f++) {
+            float t = in[f * width + b];
+            if (isfinite(t))
+                p = (1.0f - decay) * p + decay * t;
+            else
+                p *= 1.0f - decay;
+        }
+        partial[(long long)c * width + b] = p;
+    }
+}
+
+__global__ void k_moving_avg_combine(const float* __restrict__ partial,
+                                     float* __restrict__ avg,
+                                     float* __restrict__ out /* nullable */,
+                                     int width, long long frames,
+                                     int n_chunks, int chunk_frames,
+                                     float decay) {
+    int b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= width) return;
+    float a = avg[b];
+    for (int c = 0; c < n_chunks; c++) {
+        long long f0 = (long long)c * chunk_frames;
+        long long f1 = f0 + chunk_frames;
+        if (f1 > frames) f1 = frames;
+        a = powf(1.0f - decay, (float)(f1 - f0)) * a +
+            partial[(long long)c * width + b];
+    }
+    avg[b] = a;
+    if (out) out[b] = a;
+}
+
 /* ================= XlatingFir ========================================= *
  * src/blocks/xlating_fir.rs: DecimatingFir with complex band-pass taps
  * (bpf[i] = e^{i*TAU*offset/fs * i} * taps[i], :79-88) fused with the
@@ -2304,6 +2354,33 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
             r->produced = prod * f->width;
             r->status = FSDR_BOTH_SUFFICIENT;
             if (cons == 0) return FSDR_OK;
+            if (prod <= 1 &&
+                (prod == 0 || f->i_state + cons == f->history)) {
+                /* parallel fast path: chunked EMA + exact composition */
+                int cf = 64;
+                int nch = (int)((cons + cf - 1) / cf);
+                if (nch > 1024) { nch = 1024; cf = (int)((cons + nch - 1) / nch); }
+                int rc = ensure_dev(&f->d_in, &f->d_in_bytes,
+                                    (size_t)nch * f->width * 4);
+                if (rc) return rc;
+                hipLaunchKernelGGL(
+                    k_moving_avg_chunks,
+                    dim3(grid_for((long long)nch * f->width, 256)),
+                    dim3(256), 0, st, (const float*)d_in,
+                    (float*)f->d_in, (int)f->width, (long long)cons, nch,
+                    cf, f->decay);
+                HIP_TRY(hipGetLastError());
+                hipLaunchKernelGGL(k_moving_avg_combine,
+                                   dim3((unsigned)((f->width + 255) / 256)),
+                                   dim3(256), 0, st, (const float*)f->d_in,
+                                   f->d_avg,
+                                   prod ? (float*)d_out : nullptr,
+                                   (int)f->width, (long long)cons, nch, cf,
+                                   f->decay);
+                HIP_TRY(hipGetLastError());
+                f->i_state = i;
+                return FSDR_OK;
+            }
             hipLaunchKernelGGL(k_moving_avg,
                                dim3((unsigned)((f->width + 255) / 256)),
                                dim3(256), 0, st, (const float*)d_in,

@@ -746,3 +746,23 @@ def test_ring_chain_streaming(gpu, oracle_lib):
     finally:
         lib.fsdr_dev_free(d_out)
         lib.fsdr_ring_destroy(ring)
+
+
+def test_moving_avg_single_emission_fast_path(gpu, oracle_lib):
+    """history == frames triggers the parallel chunked-EMA path; the
+    chunk composition reorders f32 ops, so tolerance-compared."""
+    r = rng(151)
+    w, d = 1024, 0.1
+    frames = 1000
+    x = r.uniform(-1, 1, frames * w).astype(np.float32)
+    f = gpu.MovingAvg(w, d, frames)
+    got, c, p, s = f.filter(x, w)
+    ref, co, po, avg, i = oracle_lib.moving_avg(w, d, frames, x, w)
+    assert (c, p) == (co, po) and p == w
+    assert_close(got, ref, 1e-4)
+    # second step continues the EMA state
+    x2 = r.uniform(-1, 1, frames * w).astype(np.float32)
+    got2, c2, p2, _ = f.filter(x2, w)
+    ref2, _, _, _, _ = oracle_lib.moving_avg(w, d, frames, x2, w,
+                                             avg=avg, i_state=i)
+    assert_close(got2, ref2, 1e-4)
