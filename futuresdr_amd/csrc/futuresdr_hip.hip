@@ -1082,6 +1082,31 @@ __global__ void k_moving_avg_chunks(const float* __restrict__ in,
                                     float* __restrict__ partial, int width,
                                     long long frames, int n_chunks,
                                     int chunk_frames, float decay) {
+    /* width % 4 == 0 fast shape: one float4 of bins per thread */
+    if ((width & 3) == 0) {
+        int w4 = width >> 2;
+        long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+        long long total = (long long)n_chunks * w4;
+        long long stride = (long long)gridDim.x * blockDim.x;
+        for (; id < total; id += stride) {
+            int c = (int)(id / w4);
+            int b4 = (int)(id - (long long)c * w4);
+            long long f0 = (long long)c * chunk_frames;
+            long long f1 = f0 + chunk_frames;
+            if (f1 > frames) f1 = frames;
+            float4 p = make_float4(0.f, 0.f, 0.f, 0.f);
+            const float om = 1.0f - decay;
+            for (long long f = f0; f < f1; f++) {
+                float4 t = *(const float4*)&in[f * width + 4 * b4];
+                p.x = isfinite(t.x) ? om * p.x + decay * t.x : om * p.x;
+                p.y = isfinite(t.y) ? om * p.y + decay * t.y : om * p.y;
+                p.z = isfinite(t.z) ? om * p.z + decay * t.z : om * p.z;
+                p.w = isfinite(t.w) ? om * p.w + decay * t.w : om * p.w;
+            }
+            *(float4*)&partial[(long long)c * width + 4 * b4] = p;
+        }
+        return;
+    }
     long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
     long long total = (long long)n_chunks * width;
     long long stride = (long long)gridDim.x * blockDim.x;
@@ -2357,9 +2382,9 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
             if (prod <= 1 &&
                 (prod == 0 || f->i_state + cons == f->history)) {
                 /* parallel fast path: chunked EMA + exact composition */
-                int cf = 64;
+                int cf = 16;
                 int nch = (int)((cons + cf - 1) / cf);
-                if (nch > 1024) { nch = 1024; cf = (int)((cons + nch - 1) / nch); }
+                if (nch > 4096) { nch = 4096; cf = (int)((cons + nch - 1) / nch); }
                 int rc = ensure_dev(&f->d_in, &f->d_in_bytes,
                                     (size_t)nch * f->width * 4);
                 if (rc) return rc;
