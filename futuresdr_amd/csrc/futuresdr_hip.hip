@@ -1136,13 +1136,17 @@ __global__ void k_moving_avg_combine(const float* __restrict__ partial,
                                      float decay) {
     int b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= width) return;
+    const float om = 1.0f - decay;
+    const float full = powf(om, (float)chunk_frames); /* uniform, hoisted */
     float a = avg[b];
     for (int c = 0; c < n_chunks; c++) {
         long long f0 = (long long)c * chunk_frames;
         long long f1 = f0 + chunk_frames;
         if (f1 > frames) f1 = frames;
-        a = powf(1.0f - decay, (float)(f1 - f0)) * a +
-            partial[(long long)c * width + b];
+        float fac = (f1 - f0 == chunk_frames)
+                        ? full
+                        : powf(om, (float)(f1 - f0));
+        a = fac * a + partial[(long long)c * width + b];
     }
     avg[b] = a;
     if (out) out[b] = a;
@@ -2382,9 +2386,9 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
             if (prod <= 1 &&
                 (prod == 0 || f->i_state + cons == f->history)) {
                 /* parallel fast path: chunked EMA + exact composition */
-                int cf = 16;
+                int cf = 64;
                 int nch = (int)((cons + cf - 1) / cf);
-                if (nch > 4096) { nch = 4096; cf = (int)((cons + nch - 1) / nch); }
+                if (nch > 1024) { nch = 1024; cf = (int)((cons + nch - 1) / nch); }
                 int rc = ensure_dev(&f->d_in, &f->d_in_bytes,
                                     (size_t)nch * f->width * 4);
                 if (rc) return rc;
