@@ -2386,9 +2386,9 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
             if (prod <= 1 &&
                 (prod == 0 || f->i_state + cons == f->history)) {
                 /* parallel fast path: chunked EMA + exact composition */
-                int cf = 64;
+                int cf = 16;
                 int nch = (int)((cons + cf - 1) / cf);
-                if (nch > 1024) { nch = 1024; cf = (int)((cons + nch - 1) / nch); }
+                if (nch > 4096) { nch = 4096; cf = (int)((cons + nch - 1) / nch); }
                 int rc = ensure_dev(&f->d_in, &f->d_in_bytes,
                                     (size_t)nch * f->width * 4);
                 if (rc) return rc;
