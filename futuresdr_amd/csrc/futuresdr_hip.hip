@@ -1134,22 +1134,31 @@ __global__ void k_moving_avg_combine(const float* __restrict__ partial,
                                      int width, long long frames,
                                      int n_chunks, int chunk_frames,
                                      float decay) {
-    int b = blockIdx.x * blockDim.x + threadIdx.x;
+    /* fully parallel closed form: result[b] = om^frames * avg[b] +
+     * sum_c om^(frames - end_c) * partial[c][b]; one block per bin,
+     * chunks spread over lanes, LDS tree reduction. */
+    __shared__ float red[256];
+    const int b = blockIdx.x;
     if (b >= width) return;
     const float om = 1.0f - decay;
-    const float full = powf(om, (float)chunk_frames); /* uniform, hoisted */
-    float a = avg[b];
-    for (int c = 0; c < n_chunks; c++) {
-        long long f0 = (long long)c * chunk_frames;
-        long long f1 = f0 + chunk_frames;
-        if (f1 > frames) f1 = frames;
-        float fac = (f1 - f0 == chunk_frames)
-                        ? full
-                        : powf(om, (float)(f1 - f0));
-        a = fac * a + partial[(long long)c * width + b];
+    float acc = 0.f;
+    for (int c = threadIdx.x; c < n_chunks; c += blockDim.x) {
+        long long end = (long long)(c + 1) * chunk_frames;
+        if (end > frames) end = frames;
+        acc += powf(om, (float)(frames - end)) *
+               partial[(long long)c * width + b];
     }
-    avg[b] = a;
-    if (out) out[b] = a;
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+        if (threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        float a = powf(om, (float)frames) * avg[b] + red[0];
+        avg[b] = a;
+        if (out) out[b] = a;
+    }
 }
 
 /* ================= XlatingFir ========================================= *
@@ -2400,9 +2409,8 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                     cf, f->decay);
                 HIP_TRY(hipGetLastError());
                 hipLaunchKernelGGL(k_moving_avg_combine,
-                                   dim3((unsigned)((f->width + 255) / 256)),
-                                   dim3(256), 0, st, (const float*)f->d_in,
-                                   f->d_avg,
+                                   dim3((unsigned)f->width), dim3(256), 0,
+                                   st, (const float*)f->d_in, f->d_avg,
                                    prod ? (float*)d_out : nullptr,
                                    (int)f->width, (long long)cons, nch, cf,
                                    f->decay);
