@@ -30,9 +30,18 @@ def main():
     power = fa.wlan_moving_sum_host(
         (np.abs(sig) ** 2).astype(np.float32), 64)[:n]
     metric = fa.divide_mag_host(corr, np.maximum(power, 1e-9))
-    start = int(np.argmax(metric > 0.7))
-    print("sync-short: metric exceeds 0.7 at sample", start,
-          "(preamble starts at 300)")
+    # longest run above threshold = the STF plateau (the first few window
+    # lengths are warmup; SyncShort requires MIN_PLATEAU, sync_short.rs)
+    above = metric > 0.7
+    runs, cur, best, best_start, start = [], 0, 0, 0, 0
+    for i, a in enumerate(above):
+        cur = cur + 1 if a else 0
+        if cur == 1:
+            start = i
+        if cur > best:
+            best, best_start = cur, start
+    print(f"sync-short: longest plateau >0.7 is {best} samples starting "
+          f"at {best_start} (preamble at 300..460)")
     # sync-long: matched filter on the LTF
     taps = np.conj(ltf[::-1])
     y, _, p, _ = fa.FirCC(taps).filter(sig, n)
