@@ -54,3 +54,29 @@ def _worker(rank, world, port):
 def test_allgather_spectrum_join_gloo():
     port = 29511
     mp.spawn(_worker, args=(2, port), nprocs=2, join=True)
+
+
+def _avg_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        sys.path.insert(0, REPO)
+        import oracle as o
+        # the bench's N>1 join payload: the MovingAvg-averaged spectrum
+        mags = _channel_mags(rank)
+        avg, _, _, _, _ = o.moving_avg(FFT, 0.1, FRAMES, mags, FFT)
+        mine = torch.from_numpy(avg)
+        gathered = [torch.empty_like(mine) for _ in range(world)]
+        dist.all_gather(gathered, mine)
+        if rank == 0:
+            for rk in range(world):
+                m = _channel_mags(rk)
+                ref, _, _, _, _ = o.moving_avg(FFT, 0.1, FRAMES, m, FFT)
+                assert torch.equal(gathered[rk], torch.from_numpy(ref))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_allgather_averaged_spectrum_gloo():
+    mp.spawn(_avg_worker, args=(2, 29517), nprocs=2, join=True)
