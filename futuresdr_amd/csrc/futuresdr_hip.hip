@@ -109,7 +109,12 @@ __device__ __host__ __forceinline__ unsigned plane_floats(unsigned elems) {
  * XORing bits 5..7 of the element index into bits 2..4 makes them
  * conflict-free while keeping contiguous stages conflict-free. */
 __device__ __forceinline__ unsigned fft_swz(unsigned i) {
-    return i ^ (((i >> 5) & 7u) << 2);
+    /* full 5-bit XOR: two elements collide on a bank only when their
+     * indices differ by a multiple of 1024 — conflict-free for every
+     * Stockham stage stride at n <= 1024 (and 2-way max at 2048/4096).
+     * FFT accesses are single float2 elements, so remapping any bit of
+     * the element index is layout-legal. */
+    return i ^ ((i >> 5) & 31u);
 }
 
 __device__ __forceinline__ float2 f2_add(float2 a, float2 b) {
