@@ -656,7 +656,7 @@ __global__ __launch_bounds__(MFIR32_BLOCK) void k_fir_mfma32_tpl(
 /* forward declaration (defined with the FFT kernel below): in-block
  * 1024-pt forward Stockham FFT over swizzled LDS ping/pong buffers */
 __device__ void fft1024_block(float2* ping, float2* pong,
-                              const float2* __restrict__ twid);
+                              const float2* __restrict__ twid, int tf);
 /* result lands in `pong` (5 stages = 5 swaps); read pong[fft_swz(i)] */
 
 template <int KKD>
@@ -880,11 +880,143 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
             ping[fft_swz((unsigned)pos)] = make_float2(cre[q], cim[q]);
         }
         __syncthreads();
-        fft1024_block(ping, pong, twid); /* result in pong (swizzled) */
+        fft1024_block(ping, pong, twid, tid); /* result in pong */
         for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
             long long o = out_base + i;
             if (o < n_out) {
                 float2 v = pong[fft_swz((unsigned)i)];
+                if (out) out[o] = v; /* null = NullSink'd spectra */
+                if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+            }
+        }
+        __syncthreads();
+    }
+}
+
+template <int KKD>
+__global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv /* [4][KKD], rtv[v][u] = rt[4u+v] */,
+    long long n_out, long long n_in_valid,
+    const float2* __restrict__ twid /* 1024-entry forward table */,
+    float* __restrict__ mag_out /* nullable |X|^2 */) {
+    static_assert(true, "two 1024-pt FFT frames per tile");
+    static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
+    const unsigned elemsP = 2048 + KKD + 8;     /* per phase plane */
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    /* two phases resident at a time ([re_v0, re_v1, im_v0, im_v1]) —
+     * halves LDS vs all-phase planes, doubling resident blocks/CU;
+     * accumulators carry across the two halves, and each half's global
+     * loads are issued under the other half's MFMAs. */
+    float* planes = (float*)smem;        /* [4][SPm] */
+    float* s_rtx = planes + 4u * SPm;    /* [4][KKD+16], 15-zero prologue */
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+
+    for (int i = tid; i < 4 * (KKD + 16); i += 512) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    __syncthreads();
+
+    const unsigned span = 3 + 4 * elemsP; /* input elements per tile */
+    constexpr int NL2 =
+        (2 * (2048 + KKD + 8) + 3 + 512 - 1) / 512;
+    float2 stgA[NL2], stgB[NL2];
+    /* elements of phases {2h, 2h+1}: rel = 3 + 4i + 2h + vloc */
+    auto load_half = [&](long long tl, int h, float2 (&stg)[NL2]) {
+        const long long ib = tl * 2048 * 4;
+#pragma unroll
+        for (int j = 0; j < NL2; j++) {
+            unsigned idx = (unsigned)(tid + j * 512);
+            unsigned rel = 3 + 4 * (idx >> 1) + 2 * h + (idx & 1u);
+            long long g = ib + rel;
+            stg[j] = (rel < span && g < n_in_valid)
+                         ? in[g] : make_float2(0.f, 0.f);
+        }
+    };
+    auto write_half = [&](const float2 (&stg)[NL2]) {
+#pragma unroll
+        for (int j = 0; j < NL2; j++) {
+            unsigned idx = (unsigned)(tid + j * 512);
+            unsigned i = idx >> 1, vloc = idx & 1u;
+            if (i < elemsP) {
+                unsigned d = mfma_swz(i);
+                planes[vloc * SPm + d] = stg[j].x;
+                planes[(2 + vloc) * SPm + d] = stg[j].y;
+            }
+        }
+    };
+    const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
+#pragma unroll
+        for (int vloc = 0; vloc < 2; vloc++) {
+            const float* pre = planes + (unsigned)vloc * SPm;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm;
+            const int v = 2 * h + vloc;
+            float bfrag[KKD / 4];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++)
+                bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+            __builtin_amdgcn_s_setprio(1); /* boost MFMA waves over the
+                                              FFT/staging phases of
+                                              co-resident blocks (T5) */
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++) {
+                float a_re = pre[mfma_swz(ab + 4 * s)];
+                float a_im = pim[mfma_swz(ab + 4 * s)];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
+            }
+            __builtin_amdgcn_s_setprio(0);
+        }
+    };
+
+    load_half(blockIdx.x, 0, stgA);
+    for (long long tile = blockIdx.x;
+         tile * (long long)2048 < n_out; tile += gridDim.x) {
+        const long long out_base = tile * 2048;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        write_half(stgA);
+        __syncthreads();
+        load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
+        mfma_half(0, cre, cim);
+        __syncthreads();
+        write_half(stgB);
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)2048 < n_out)
+            load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
+        mfma_half(1, cre, cim);
+        __syncthreads(); /* phase planes are dead; reuse them as FFT LDS:
+                            frame f ping = fbase + f*2048, pong = +1024 */
+        float2* fbase = (float2*)planes; /* 4096 float2 = 32 KB */
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            int pos = wave * 256 + 16 * row + r16; /* y2 index in tile */
+            int fr = pos >> 10, idx = pos & 1023;
+            fbase[fr * 2048 + fft_swz((unsigned)idx)] =
+                make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+        {
+            int fl = tid >> 8, tf = tid & 255;
+            fft1024_block(fbase + fl * 2048, fbase + fl * 2048 + 1024,
+                          twid, tf);
+        }
+        for (int i = tid; i < 2048; i += 512) {
+            long long o = out_base + i;
+            if (o < n_out) {
+                int fr = i >> 10, idx = i & 1023;
+                float2 v = fbase[fr * 2048 + 1024 + fft_swz((unsigned)idx)];
                 if (out) out[o] = v; /* null = NullSink'd spectra */
                 if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
             }
@@ -1424,9 +1556,9 @@ __device__ __forceinline__ float2 cmul_tw(float2 a, float2 w, int inverse) {
  * chain's fused decim+FFT kernel. Same radix-4 DIF scheme and fft_swz
  * LDS swizzle as k_fft_stockham. */
 __device__ void fft1024_block(float2* ping, float2* pong,
-                              const float2* __restrict__ twid) {
+                              const float2* __restrict__ twid,
+                              int tf /* thread-in-frame, stride 256 */) {
     const int n = 1024;
-    const int tf = threadIdx.x;
     float2* a = ping;
     float2* b = pong;
     int scur = 1;
@@ -2837,6 +2969,23 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
         float2* spec_dst = (float2*)d_out; /* null + mag-only: skip the
                                               discarded spectra write */
         if (!spec_dst && !d_mag) spec_dst = out2;
+        const char* b512 = getenv("FSDR_CHAIN_BLOCK512");
+        if (b512 && atoi(b512) != 0 && prod % 2048 == 0) {
+            long long tiles2 = (long long)prod / 2048;
+            int grid2 = (int)std::min<long long>(tiles2, cap);
+            unsigned eP2 = 2048 + 80 + 8;
+            size_t lds2 = (4 * (size_t)((eP2 + 31u) & ~31u) + 4 * (80 + 16))
+                          * sizeof(float);
+            hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma2_tpl<80>),
+                               dim3(grid2), dim3(512), lds2, st,
+                               (const float2*)d_in, spec_dst,
+                               c->fused->d_mtaps, (long long)prod,
+                               (long long)n_in,
+                               (const float2*)c->fft->d_twid,
+                               (float*)d_mag);
+            HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
         hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_tpl<80>),
                            dim3(grid), dim3(MDFIR_BLOCK), lds, st,
                            (const float2*)d_in, spec_dst, c->fused->d_mtaps,
