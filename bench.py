@@ -46,7 +46,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--samples", type=int, default=1 << 26,
+    p.add_argument("--samples", type=int, default=1 << 28,
                    help="chain-input Complex32 samples per step per GPU")
     p.add_argument("--fft", type=int, default=1024)
     p.add_argument("--decim", type=int, default=4)
