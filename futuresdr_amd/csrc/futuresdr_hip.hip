@@ -1308,7 +1308,7 @@ __global__ void k_moving_avg_combine(const float* __restrict__ partial,
 __global__ __launch_bounds__(256) void k_xlating_decim_ccf32(
     const float2* __restrict__ in, float2* __restrict__ out,
     const float2* __restrict__ taps /* reversed bpf taps */, int n_taps,
-    long long decim, long long n_out, long long n_in_valid, float theta,
+    long long decim, long long n_out, long long n_in_valid, double theta,
     float p0r, float p0i) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float2* s_h = (float2*)smem;
@@ -1328,8 +1328,11 @@ __global__ __launch_bounds__(256) void k_xlating_decim_ccf32(
             sim = fmaf(x.x, hh.y, sim);
             sim = fmaf(x.y, hh.x, sim);
         }
+        /* f64 angle + mod-2pi reduction (see k_rotator) */
+        double a = theta * (double)(k + 1);
+        a -= 6.283185307179586 * floor(a * 0.15915494309189535);
         float s, cth;
-        __sincosf(theta * (float)(k + 1), &s, &cth);
+        __sincosf((float)a, &s, &cth);
         float pr = cth * p0r - s * p0i;
         float pi = cth * p0i + s * p0r;
         out[k] = make_float2(sre * pr - sim * pi, sre * pi + sim * pr);
@@ -1460,12 +1463,16 @@ __global__ void k_moving_avg(const float* __restrict__ in,
  * oracle's own drift (documented in tests). */
 __global__ void k_rotator(const float2* __restrict__ in,
                           float2* __restrict__ out, long long n,
-                          float theta, float p0r, float p0i) {
+                          double theta, float p0r, float p0i) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
          i < n; i += stride) {
+        /* angle product + mod-2pi reduction in f64: f32 theta*(i+1) loses
+         * ~0.25 rad past i~3e6 and (float)(i+1) is inexact past 2^24 */
+        double a = theta * (double)(i + 1);
+        a -= 6.283185307179586 * floor(a * 0.15915494309189535);
         float s, c;
-        __sincosf(theta * (float)(i + 1), &s, &c);
+        __sincosf((float)a, &s, &c);
         float pr = c * p0r - s * p0i;
         float pi = c * p0i + s * p0r;
         float2 x = in[i];
@@ -1828,6 +1835,11 @@ struct fsdr_filter {
     void* d_out = nullptr;
     size_t d_in_bytes = 0, d_out_bytes = 0;
     size_t item_in = 8, item_out = 8;
+    /* dedicated kernel scratch (MovingAvg chunk partials). MUST be
+     * distinct from d_in: on the fsdr_filter_host path d_in holds the
+     * staged input while the partials kernel is still reading it. */
+    void* d_scratch = nullptr;
+    size_t d_scratch_bytes = 0;
 };
 
 static int ensure_dev(void** p, size_t* cur, size_t want) {
@@ -2227,6 +2239,7 @@ extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (f->d_twid) (void)hipFree(f->d_twid);
     if (f->d_in) (void)hipFree(f->d_in);
     if (f->d_out) (void)hipFree(f->d_out);
+    if (f->d_scratch) (void)hipFree(f->d_scratch);
     delete f;
 }
 
@@ -2535,19 +2548,19 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                 int cf = 64; /* measured best (16 and 256 both slower) */
                 int nch = (int)((cons + cf - 1) / cf);
                 if (nch > 1024) { nch = 1024; cf = (int)((cons + nch - 1) / nch); }
-                int rc = ensure_dev(&f->d_in, &f->d_in_bytes,
+                int rc = ensure_dev(&f->d_scratch, &f->d_scratch_bytes,
                                     (size_t)nch * f->width * 4);
                 if (rc) return rc;
                 hipLaunchKernelGGL(
                     k_moving_avg_chunks,
                     dim3(grid_for((long long)nch * f->width, 256)),
                     dim3(256), 0, st, (const float*)d_in,
-                    (float*)f->d_in, (int)f->width, (long long)cons, nch,
-                    cf, f->decay);
+                    (float*)f->d_scratch, (int)f->width, (long long)cons,
+                    nch, cf, f->decay);
                 HIP_TRY(hipGetLastError());
                 hipLaunchKernelGGL(k_moving_avg_combine,
                                    dim3((unsigned)f->width), dim3(256), 0,
-                                   st, (const float*)f->d_in, f->d_avg,
+                                   st, (const float*)f->d_scratch, f->d_avg,
                                    prod ? (float*)d_out : nullptr,
                                    (int)f->width, (long long)cons, nch, cf,
                                    f->decay);
