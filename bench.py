@@ -2,8 +2,14 @@
 """bench.py — BASELINE.json metric: MSample/s of Complex32 input through
 the 127-tap FIR -> decim-4 -> 1024-pt FFT chain, on N GPUs of one node.
 
-A step = one pass of the chain over one HBM-resident batch of synthetic
-samples (re/im iid uniform[-1,1), device-generated, seeded). At N>1 each
+A step = SOURCE WRITE + one pass of the chain over one HBM-resident batch
+of synthetic samples (re/im iid uniform[-1,1), device-generated, seeded).
+The source's buffer write (the NullSource analogue,
+src/blocks/null_source.rs:53-66) is INSIDE the timed region: every step
+re-fills the input batch on-device before the chain consumes it. A
+separate streaming leg (reported as "streaming" in the JSON, never as
+`value`) pushes host chunks through the pinned fsdr_ring (H2D on the copy
+stream overlapped with compute) — the PCIe-fed rate. At N>1 each
 rank runs an independent channel on its own GPU (config 4) and the
 spectrum-combine step joins the ranks' spectra with one RCCL all-gather
 per step. The payload is the reference spectrum SINK's output — the
@@ -55,6 +61,12 @@ def parse_args():
                    help="all-gather every raw magnitude frame instead of "
                         "the MovingAvg-averaged spectrum")
     p.add_argument("--skip-cpu-baseline", action="store_true")
+    p.add_argument("--skip-streaming", action="store_true",
+                   help="skip the PCIe-fed ring streaming leg")
+    p.add_argument("--streaming-chunk", type=int, default=1 << 22,
+                   help="ring chunk size in samples for the streaming leg")
+    p.add_argument("--streaming-chunks", type=int, default=64,
+                   help="number of chunks for the streaming leg")
     p.add_argument("--cpu-sample", type=int, default=0,
                    help="fixed CPU-baseline sample size (0 = auto ~10s)")
     p.add_argument("--traffic-file", default=os.path.join(
@@ -119,6 +131,66 @@ def measure_chain_roofline(fa, torch, chain, d_in, n_samples, taps1,
     }
 
 
+def measure_streaming(fa, lib, chain, taps1, taps2, decim, fft_len,
+                      chunk, n_chunks):
+    """PCIe-fed leg: host chunks -> pinned fsdr_ring (async H2D on the
+    ring's copy stream, overlapped with compute) -> fused chain kernel ->
+    MovingAvg mag sink on-device. The host source cost (filling the
+    pinned slice, the NullSource analogue) is inside the timed region.
+    Reported beside the resident number; never `value`."""
+    g_len = taps1.size + taps2.size - 1
+    reserved = g_len - 1 + decim * fft_len + decim
+    ring = lib.fsdr_ring_create(4, chunk, 8, reserved)
+    if not ring:
+        raise RuntimeError(lib.fsdr_last_error().decode())
+    out_cap = (reserved + chunk) // decim + fft_len
+    d_mag = ctypes.c_void_p()
+    if lib.fsdr_dev_alloc(ctypes.byref(d_mag), out_cap * 4) != 0:
+        raise RuntimeError(lib.fsdr_last_error().decode())
+    import torch
+    st = torch.cuda.current_stream()
+    hp = ctypes.c_void_p()
+    items = ctypes.c_size_t()
+    dp = ctypes.c_void_p()
+    got = ctypes.c_size_t()
+
+    def push(n):
+        total = 0
+        for _ in range(n):
+            lib.fsdr_ring_writer_acquire(ring, ctypes.byref(hp),
+                                         ctypes.byref(items))
+            ctypes.memset(hp, 0, chunk * 8)  # the source's buffer write
+            lib.fsdr_ring_writer_commit(ring, chunk)
+            lib.fsdr_ring_reader_acquire(ring, ctypes.byref(dp),
+                                         ctypes.byref(got))
+            cons, _prod = chain.run_dev(dp.value, got.value, 0, 0,
+                                        d_mag.value, out_cap,
+                                        stream=st.cuda_stream)
+            lib.fsdr_ring_reader_release_consumed(
+                ring, cons, ctypes.c_void_p(st.cuda_stream))
+            total += chunk
+        return total
+    try:
+        push(4)  # warmup
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        n = push(n_chunks)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        return {
+            "value": round(n / dt / 1e6, 2),
+            "unit": "MSample/s",
+            "chunk_samples": chunk,
+            "chunks": n_chunks,
+            "note": "host-fed pinned ring (PCIe-inclusive, host memset "
+                    "source in timed region); not comparable to `value` "
+                    "(HBM-resident)",
+        }
+    finally:
+        lib.fsdr_dev_free(d_mag)
+        lib.fsdr_ring_destroy(ring)
+
+
 def measure_cpu_baseline(taps1, taps2, decim, fft_len, fixed_sample):
     """oracle chain (the CPU restatement, kind 'port') on all host cores,
     bounded to ~10 s of work."""
@@ -169,11 +241,20 @@ def main():
     lib = fa.lib()
     st = torch.cuda.current_stream()
 
+    # Initialize RCCL whenever launched under torchrun (even world=1) so
+    # the distributed path — nccl init, all_gather_into_tensor, barrier,
+    # MAX-reduce — is exercised on hardware by single-GPU runs too.
     td = None
-    if world > 1:
+    if world > 1 or "TORCHELASTIC_RUN_ID" in os.environ \
+            or os.environ.get("FSDR_FORCE_DIST"):
         import torch.distributed as td_
         td = td_
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29871")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
         td.init_process_group("nccl")
+        log(f"torch.distributed initialized: rank {rank}/{world} (nccl)")
 
     beta = fa.kaiser_beta(1e-4)
     taps1 = fa.lowpass_kaiser_n(args.taps, beta, 0.1)
@@ -195,12 +276,19 @@ def main():
     avg_sink = fa.MovingAvg(args.fft, 0.1, frames)
     avg_spec = torch.empty(args.fft, dtype=torch.float32, device="cuda")
     gathered = None
-    if world > 1:
+    if td is not None:
         payload = prod if args.gather_frames else args.fft
         gathered = torch.empty(world * payload, dtype=torch.float32,
                                device="cuda")
 
+    seed_ctr = [0]
+
     def step():
+        # the SOURCE: refill the input batch on-device (NullSource writes
+        # its whole buffer every call — null_source.rs:53-66)
+        seed_ctr[0] += 1
+        fa.fill_uniform_dev(d_in.value, S, seed=seed_ctr[0] * 131 + rank,
+                            stream=st.cuda_stream)
         chain.run_dev(d_in.value, S, 0, 0, mag.data_ptr(), prod,
                       stream=st.cuda_stream)
         avg_sink.filter_dev(mag.data_ptr(), prod, avg_spec.data_ptr(),
@@ -240,6 +328,13 @@ def main():
             log("measuring CPU baseline (oracle chain, all cores)...")
             cpu_baseline = measure_cpu_baseline(taps1, taps2, args.decim,
                                                 args.fft, args.cpu_sample)
+        streaming = None
+        if n_gpus == 1 and not args.skip_streaming:
+            log("measuring PCIe-fed ring streaming leg...")
+            streaming = measure_streaming(fa, lib, chain, taps1, taps2,
+                                          args.decim, args.fft,
+                                          args.streaming_chunk,
+                                          args.streaming_chunks)
         result = {
             "metric": "MSample/s through 127-tap C32 FIR→decim4→"
                       "1k-FFT flowgraph @1/2/4/8 GPU",
@@ -256,6 +351,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "workload": "fir127_decim4_fft1024",
+                "source_in_timed_region": True,
                 "taps1": int(taps1.size),
                 "taps2": int(taps2.size),
                 "decim": args.decim,
@@ -271,6 +367,7 @@ def main():
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
+            "streaming": streaming,
         }
         print(json.dumps(result), flush=True)
 

@@ -184,6 +184,8 @@ def _load():
                                              ctypes.POINTER(sz)]
     lib.fsdr_ring_reader_release.restype = ctypes.c_int
     lib.fsdr_ring_reader_release.argtypes = [vp]
+    lib.fsdr_ring_reader_release_consumed.restype = ctypes.c_int
+    lib.fsdr_ring_reader_release_consumed.argtypes = [vp, sz, vp]
     lib.fsdr_ring_destroy.argtypes = [vp]
     _lib = lib
     return lib

@@ -3030,19 +3030,30 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
 /* ================= ring (Slab-style) ================================== */
 
 struct RingBuf {
-    void* host = nullptr;  /* pinned, reserved + items */
+    void* host = nullptr;  /* pinned; carry capacity + payload */
     void* dev = nullptr;
-    size_t items = 0;      /* valid payload items (excl. reserved prefix) */
-    hipEvent_t ev = nullptr;
+    size_t items = 0;      /* valid payload items (after the carry region) */
+    hipEvent_t ev = nullptr;      /* copy (H2D [+ carry D2D]) complete */
+    hipEvent_t free_ev = nullptr; /* consumer done reading dev */
 };
 
+/* Slab-exact streaming ring (slab.rs:110-152,369-399): the reader reports
+ * how many items it CONSUMED and the ring carries the unconsumed
+ * remainder in front of the next buffer — device-side (one small D2D on
+ * the copy stream), so arbitrary chunk sizes stream exactly like the
+ * reference block leaving leftovers in the slab buffer. `reserved`
+ * is the carry CAPACITY (>= the consumer's worst-case leftover; for a
+ * FIR chain that is taps-1 + decim*fft_len). */
 struct fsdr_ring {
     size_t n_buffers, items_per_buffer, item_bytes, reserved;
     std::vector<RingBuf> bufs;
     std::deque<int> empty_q, full_q;
     int writer_cur = -1, reader_cur = -1;
-    std::vector<char> tail; /* last `reserved` items of previous buffer */
-    bool tail_valid = false;
+    int prev_buf = -1;          /* released buffer still holding the tail */
+    size_t prev_tail_off = 0;   /* byte offset of the tail in prev dev buf */
+    size_t carry_items = 0;     /* items to prepend at the next acquire */
+    size_t presented_items = 0; /* what the current reader_acquire returned */
+    size_t presented_off = 0;   /* byte offset of presented span start */
     std::mutex mu;
     std::condition_variable cv;
     hipStream_t copy_stream = nullptr;
@@ -3062,8 +3073,6 @@ extern "C" fsdr_ring* fsdr_ring_create(size_t n_buffers,
     r->items_per_buffer = items_per_buffer;
     r->item_bytes = item_bytes;
     r->reserved = reserved_items;
-    r->tail.resize(reserved_items * item_bytes, 0);
-    r->tail_valid = true; /* zero history = stream start */
     if (hipStreamCreate(&r->copy_stream) != hipSuccess) {
         set_err("stream create failed");
         delete r;
@@ -3074,8 +3083,10 @@ extern "C" fsdr_ring* fsdr_ring_create(size_t n_buffers,
         RingBuf b;
         if (hipHostMalloc(&b.host, bytes) != hipSuccess ||
             hipMalloc(&b.dev, bytes) != hipSuccess ||
-            hipEventCreate(&b.ev) != hipSuccess) {
+            hipEventCreate(&b.ev) != hipSuccess ||
+            hipEventCreate(&b.free_ev) != hipSuccess) {
             set_err("ring buffer alloc failed");
+            r->bufs.push_back(b);
             fsdr_ring_destroy(r);
             return nullptr;
         }
@@ -3102,26 +3113,13 @@ extern "C" int fsdr_ring_writer_commit(fsdr_ring* r, size_t items) {
     if (!r || r->writer_cur < 0) return FSDR_ERR_INVALID;
     RingBuf& b = r->bufs[r->writer_cur];
     b.items = items;
-    /* history prefix: tail of the previous committed buffer, exactly the
-     * slab reserved-prefix merge (slab.rs:369-399) */
-    memcpy(b.host, r->tail.data(), r->reserved * r->item_bytes);
-    if (r->reserved) {
-        size_t ib = r->item_bytes;
-        if (items >= r->reserved) {
-            memcpy(r->tail.data(),
-                   (char*)b.host + (r->reserved + items - r->reserved) * ib,
-                   r->reserved * ib);
-        } else {
-            /* shift old tail, append new items */
-            memmove(r->tail.data(), r->tail.data() + items * ib,
-                    (r->reserved - items) * ib);
-            memcpy(r->tail.data() + (r->reserved - items) * ib,
-                   (char*)b.host + r->reserved * ib, items * ib);
-        }
-    }
-    HIP_TRY(hipMemcpyAsync(b.dev, b.host,
-                           (r->reserved + items) * r->item_bytes,
-                           hipMemcpyHostToDevice, r->copy_stream));
+    size_t off = r->reserved * r->item_bytes;
+    /* don't overwrite dev while the consumer may still read it (a
+     * never-recorded free_ev makes this wait a no-op) */
+    HIP_TRY(hipStreamWaitEvent(r->copy_stream, b.free_ev, 0));
+    HIP_TRY(hipMemcpyAsync((char*)b.dev + off, (char*)b.host + off,
+                           items * r->item_bytes, hipMemcpyHostToDevice,
+                           r->copy_stream));
     HIP_TRY(hipEventRecord(b.ev, r->copy_stream));
     {
         std::lock_guard<std::mutex> lk(r->mu);
@@ -3141,21 +3139,64 @@ extern "C" int fsdr_ring_reader_acquire(fsdr_ring* r, void** dev_ptr,
     r->full_q.pop_front();
     lk.unlock();
     RingBuf& b = r->bufs[r->reader_cur];
+    size_t ib = r->item_bytes;
+    size_t off = (r->reserved - r->carry_items) * ib;
+    if (r->carry_items > 0 && r->prev_buf >= 0) {
+        /* prepend the previous buffer's unconsumed tail (slab.rs:369-399),
+         * after the consumer's reads of it have drained */
+        RingBuf& pb = r->bufs[r->prev_buf];
+        HIP_TRY(hipStreamWaitEvent(r->copy_stream, pb.free_ev, 0));
+        HIP_TRY(hipMemcpyAsync((char*)b.dev + off,
+                               (char*)pb.dev + r->prev_tail_off,
+                               r->carry_items * ib,
+                               hipMemcpyDeviceToDevice, r->copy_stream));
+        HIP_TRY(hipEventRecord(b.ev, r->copy_stream));
+    }
+    if (r->prev_buf >= 0) {
+        std::lock_guard<std::mutex> lk2(r->mu);
+        r->empty_q.push_back(r->prev_buf);
+        r->prev_buf = -1;
+        r->cv.notify_all();
+    }
     HIP_TRY(hipEventSynchronize(b.ev));
-    *dev_ptr = b.dev; /* includes the reserved-history prefix */
-    *items = b.items; /* payload items after the prefix */
+    r->presented_items = r->carry_items + b.items;
+    r->presented_off = off;
+    r->carry_items = 0;
+    *dev_ptr = (char*)b.dev + off;
+    *items = r->presented_items;
+    return FSDR_OK;
+}
+
+/* Reader reports how much of the presented span it consumed; the
+ * remainder is carried in front of the next buffer. `stream` is the
+ * consumer's compute stream (the carry copy and buffer reuse are ordered
+ * after work already enqueued there); NULL = default stream. */
+extern "C" int fsdr_ring_reader_release_consumed(fsdr_ring* r,
+                                                 size_t consumed,
+                                                 void* stream) {
+    if (!r || r->reader_cur < 0) return FSDR_ERR_INVALID;
+    if (consumed > r->presented_items) {
+        set_err("release_consumed: consumed > presented");
+        return FSDR_ERR_INVALID;
+    }
+    size_t un = r->presented_items - consumed;
+    if (un > r->reserved) {
+        set_err("release_consumed: unconsumed tail exceeds ring "
+                "reserved capacity");
+        return FSDR_ERR_INVALID;
+    }
+    RingBuf& b = r->bufs[r->reader_cur];
+    HIP_TRY(hipEventRecord(b.free_ev, (hipStream_t)stream));
+    r->prev_buf = r->reader_cur;
+    r->prev_tail_off = r->presented_off + consumed * r->item_bytes;
+    r->carry_items = un;
+    r->reader_cur = -1;
     return FSDR_OK;
 }
 
 extern "C" int fsdr_ring_reader_release(fsdr_ring* r) {
-    if (!r || r->reader_cur < 0) return FSDR_ERR_INVALID;
-    {
-        std::lock_guard<std::mutex> lk(r->mu);
-        r->empty_q.push_back(r->reader_cur);
-        r->reader_cur = -1;
-    }
-    r->cv.notify_all();
-    return FSDR_OK;
+    return fsdr_ring_reader_release_consumed(r, r ? r->presented_items : 0,
+                                             nullptr);
 }
 
 extern "C" void fsdr_ring_destroy(fsdr_ring* r) {
@@ -3164,6 +3205,7 @@ extern "C" void fsdr_ring_destroy(fsdr_ring* r) {
         if (b.host) (void)hipHostFree(b.host);
         if (b.dev) (void)hipFree(b.dev);
         if (b.ev) (void)hipEventDestroy(b.ev);
+        if (b.free_ev) (void)hipEventDestroy(b.free_ev);
     }
     if (r->copy_stream) (void)hipStreamDestroy(r->copy_stream);
     delete r;

@@ -218,20 +218,25 @@ void fsdr_chain_destroy(fsdr_chain* c);
 
 /* ---- Ring (Slab-style stream buffer) ---------------------------------- *
  * N pinned-host buffers + N device mirrors circulating between an empty
- * queue (writer side) and a full queue (reader side), with a
- * reserved_items history prefix maintained exactly like slab.rs:369-399:
- * the first `reserved` items of each acquired read-slice are the tail of
- * the previous buffer. Writer: acquire -> fill host slice -> commit(n)
- * (enqueues async H2D on the ring's copy stream). Reader: acquire (waits
- * for the H2D event; yields the device pointer including history prefix)
- * -> ... launch kernels ... -> release (recycles the buffer, copies the
- * history tail). Single-producer single-consumer. */
+ * queue (writer side) and a full queue (reader side), carrying the
+ * reader's UNCONSUMED tail in front of the next buffer exactly like
+ * slab.rs:369-399 (device-side D2D on the ring's copy stream), so
+ * arbitrary chunk sizes stream exactly. reserved_items = carry capacity
+ * (>= the consumer's worst-case leftover; for a FIR chain:
+ * taps-1 + decim*fft_len + decim). Writer: acquire -> fill host slice ->
+ * commit(n) (enqueues async H2D on the copy stream). Reader: acquire
+ * (waits for the copies; yields the device pointer at the carry start
+ * and items = carry + payload) -> ... launch kernels consuming
+ * `consumed` items ... -> release_consumed(consumed, compute_stream).
+ * release() = consumed everything. Single-producer single-consumer. */
 typedef struct fsdr_ring fsdr_ring;
 fsdr_ring* fsdr_ring_create(size_t n_buffers, size_t items_per_buffer,
                             size_t item_bytes, size_t reserved_items);
 int  fsdr_ring_writer_acquire(fsdr_ring* r, void** host_ptr, size_t* items);
 int  fsdr_ring_writer_commit(fsdr_ring* r, size_t items);
 int  fsdr_ring_reader_acquire(fsdr_ring* r, void** dev_ptr, size_t* items);
+int  fsdr_ring_reader_release_consumed(fsdr_ring* r, size_t consumed,
+                                       void* stream);
 int  fsdr_ring_reader_release(fsdr_ring* r);
 void fsdr_ring_destroy(fsdr_ring* r);
 

@@ -395,57 +395,66 @@ def test_chain_full_size_properties(gpu):
 
 # ---------------- ring (Slab-style) ------------------------------------
 
+def _stream_through_ring(gpu, lib, chunks, reserved, consume):
+    """Push `chunks` (list of complex64 arrays) through the ring; call
+    `consume(dev_ptr, n_presented) -> (consumed, output_array)` per
+    acquire; release_consumed carries the leftover slab-style. Returns
+    the concatenated outputs."""
+    max_chunk = max(c.size for c in chunks)
+    ring = lib.fsdr_ring_create(4, max_chunk, 8, reserved)
+    assert ring, lib.fsdr_last_error().decode()
+    outs = []
+    try:
+        for x in chunks:
+            hp = ctypes.c_void_p()
+            items = ctypes.c_size_t()
+            assert lib.fsdr_ring_writer_acquire(
+                ring, ctypes.byref(hp), ctypes.byref(items)) == 0
+            assert items.value >= x.size
+            ctypes.memmove(hp, ctypes.c_void_p(x.ctypes.data), x.size * 8)
+            assert lib.fsdr_ring_writer_commit(ring, x.size) == 0
+            dp = ctypes.c_void_p()
+            got_items = ctypes.c_size_t()
+            assert lib.fsdr_ring_reader_acquire(
+                ring, ctypes.byref(dp), ctypes.byref(got_items)) == 0
+            cons, out = consume(dp.value, got_items.value)
+            outs.append(out)
+            assert lib.fsdr_ring_reader_release_consumed(
+                ring, cons, None) == 0, lib.fsdr_last_error().decode()
+    finally:
+        lib.fsdr_ring_destroy(ring)
+    return np.concatenate(outs)
+
+
 def test_ring_history_prefix(gpu, oracle_lib):
     """Stream a long signal through the pinned ring in chunks and run the
-    FIR on each acquired device buffer; the reserved prefix must carry
-    exactly taps-1 samples of history so the concatenated outputs equal
-    the single-shot oracle (slab.rs reserved-prefix semantics)."""
+    FIR on each acquired device buffer; the carried unconsumed tail
+    (slab.rs:369-399 semantics) makes the concatenated outputs equal the
+    single-shot oracle on the raw stream — no zero prologue, exactly like
+    the reference block consuming less on its first work()."""
     lib = gpu.lib()
     r = rng(67)
     taps = r.uniform(-1, 1, 127).astype(np.float32)
     n_total, chunk = 40960, 4096
     x = cplx(r, n_total)
-    reserved = 126
-    ring = lib.fsdr_ring_create(4, chunk, 8, reserved)
-    assert ring
+    fir = gpu.Fir(taps)
+    d_out = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), 2 * chunk * 8) == 0
     try:
-        fir = gpu.Fir(taps)
-        d_out = ctypes.c_void_p()
-        assert lib.fsdr_dev_alloc(ctypes.byref(d_out), chunk * 8) == 0
-        outs = []
-        try:
-            for off in range(0, n_total, chunk):
-                hp = ctypes.c_void_p()
-                items = ctypes.c_size_t()
-                assert lib.fsdr_ring_writer_acquire(
-                    ring, ctypes.byref(hp), ctypes.byref(items)) == 0
-                n = min(chunk, n_total - off)
-                ctypes.memmove(hp, ctypes.c_void_p(
-                    x[off:off + n].ctypes.data), n * 8)
-                assert lib.fsdr_ring_writer_commit(ring, n) == 0
-                dp = ctypes.c_void_p()
-                got_items = ctypes.c_size_t()
-                assert lib.fsdr_ring_reader_acquire(
-                    ring, ctypes.byref(dp), ctypes.byref(got_items)) == 0
-                n_in = reserved + got_items.value
-                c, p, s = fir.filter_dev(dp.value, n_in, d_out.value, chunk)
-                gpu.synchronize()
-                h = np.zeros(p, np.complex64)
-                lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
-                                    p * 8)
-                outs.append(h)
-                assert lib.fsdr_ring_reader_release(ring) == 0
-        finally:
-            lib.fsdr_dev_free(d_out)
-        got = np.concatenate(outs)
-        # first chunk sees a zero history prefix; the oracle equivalent is
-        # the signal with 126 zeros prepended
-        xz = np.concatenate([np.zeros(reserved, np.complex64), x])
-        ref, c, p, s = oracle_lib.fir_cf32(taps, xz, got.size)
+        def consume(dp, n_in):
+            c, p, s = fir.filter_dev(dp, n_in, d_out.value, 2 * chunk)
+            gpu.synchronize()
+            h = np.zeros(p, np.complex64)
+            lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
+                                p * 8)
+            return c, h
+        chunks = [x[o:o + chunk] for o in range(0, n_total, chunk)]
+        got = _stream_through_ring(gpu, lib, chunks, taps.size - 1, consume)
+        ref, c, p, s = oracle_lib.fir_cf32(taps, x, n_total)
         assert p == got.size
-        assert_close(got, ref)
+        assert_close(got, ref[:got.size])
     finally:
-        lib.fsdr_ring_destroy(ring)
+        lib.fsdr_dev_free(d_out)
 
 
 # ---------------- dev path on a torch stream ---------------------------
@@ -691,61 +700,67 @@ def test_divide_mag_parity(gpu):
     assert_close(got, ref, 1e-5)
 
 
-def test_ring_chain_streaming(gpu, oracle_lib):
-    """The production streaming path: host chunks flow through the
-    pinned-host ring (Slab semantics, 252-sample history prefix) into the
-    fused chain kernel; concatenated spectra must equal the one-shot
-    oracle chain on the zero-history-prefixed stream. Chunk size is a
-    multiple of 4*fft_len so every chunk's decimated output is whole
-    frames (the chain consumes each chunk fully, like the reference block
-    leaves no aligned leftovers)."""
+def _chain_stream_case(gpu, oracle_lib, fft_len, chunk_sizes, seed,
+                       decim=4):
+    """Stream arbitrary chunk sizes through ring -> fused chain; the
+    carried unconsumed tail (input remainder AND non-frame-aligned
+    decimated leftovers, which stay upstream as unconsumed input) makes
+    the concatenated spectra equal the one-shot oracle chain."""
     lib = gpu.lib()
-    r = rng(149)
+    r = rng(seed)
     t1 = r.uniform(-1, 1, 127).astype(np.float32)
     t2 = r.uniform(-1, 1, 127).astype(np.float32)
-    fft_len = 256
-    chunk = 4 * fft_len * 16            # 16384 samples/chunk
-    n_total = chunk * 8
+    n_total = int(sum(chunk_sizes))
     x = cplx(r, n_total)
-    g_len = t1.size + t2.size - 1       # fused taps
-    reserved = g_len - 1                # 252
-    ring = lib.fsdr_ring_create(4, chunk, 8, reserved)
-    assert ring
-    chain = gpu.Chain(t1, t2, 4, fft_len)
+    g_len = t1.size + t2.size - 1
+    # worst-case leftover: g_len-1 window history + an unfilled frame
+    reserved = g_len - 1 + decim * fft_len + decim
+    chain = gpu.Chain(t1, t2, decim, fft_len)
+    max_chunk = max(chunk_sizes)
+    out_cap = (reserved + max_chunk) // decim + fft_len
     d_out = ctypes.c_void_p()
-    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), chunk * 8) == 0
-    outs = []
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), out_cap * 8) == 0
     try:
-        for off in range(0, n_total, chunk):
-            hp = ctypes.c_void_p()
-            items = ctypes.c_size_t()
-            assert lib.fsdr_ring_writer_acquire(
-                ring, ctypes.byref(hp), ctypes.byref(items)) == 0
-            ctypes.memmove(hp, ctypes.c_void_p(
-                x[off:off + chunk].ctypes.data), chunk * 8)
-            assert lib.fsdr_ring_writer_commit(ring, chunk) == 0
-            dp = ctypes.c_void_p()
-            got_items = ctypes.c_size_t()
-            assert lib.fsdr_ring_reader_acquire(
-                ring, ctypes.byref(dp), ctypes.byref(got_items)) == 0
-            n_in = reserved + got_items.value
-            cons, prod = chain.run_dev(dp.value, n_in, d_out.value, chunk)
+        def consume(dp, n_in):
+            cons, prod = chain.run_dev(dp, n_in, d_out.value, out_cap)
             gpu.synchronize()
-            assert cons == got_items.value  # whole chunk consumed
             h = np.zeros(prod, np.complex64)
             lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
                                 prod * 8)
-            outs.append(h)
-            assert lib.fsdr_ring_reader_release(ring) == 0
-        got = np.concatenate(outs)
-        xz = np.concatenate([np.zeros(reserved, np.complex64), x])
-        ref, _ = oracle_lib.chain_cf32(t1, t2, 4, fft_len, xz)
-        assert got.size == ref.size
+            return cons, h
+        offs = np.concatenate([[0], np.cumsum(chunk_sizes)]).astype(int)
+        chunks = [x[offs[i]:offs[i + 1]] for i in range(len(chunk_sizes))]
+        got = _stream_through_ring(gpu, lib, chunks, reserved, consume)
+        ref, _ = oracle_lib.chain_cf32(t1, t2, decim, fft_len, x)
+        assert got.size == ref.size, (got.size, ref.size)
         rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
         assert rel < 2e-4, rel
     finally:
         lib.fsdr_dev_free(d_out)
-        lib.fsdr_ring_destroy(ring)
+
+
+def test_ring_chain_streaming(gpu, oracle_lib):
+    """Production streaming path at an aligned chunk size (every chunk's
+    decimated output is whole frames)."""
+    fft_len = 256
+    chunk = 4 * fft_len * 16
+    _chain_stream_case(gpu, oracle_lib, fft_len, [chunk] * 8, 149)
+
+
+def test_ring_chain_streaming_unaligned(gpu, oracle_lib):
+    """Arbitrary (random) chunk sizes: non-frame-aligned leftovers must
+    carry across fsdr_chain_run_dev calls like slab does."""
+    r = rng(151)
+    sizes = r.integers(257, 20000, size=12).tolist()
+    _chain_stream_case(gpu, oracle_lib, 256, sizes, 153)
+
+
+def test_ring_chain_streaming_tiny_chunks(gpu, oracle_lib):
+    """Chunks smaller than the filter window: several acquires produce
+    nothing until enough history accumulates."""
+    r = rng(157)
+    sizes = r.integers(40, 900, size=30).tolist()
+    _chain_stream_case(gpu, oracle_lib, 64, sizes, 159)
 
 
 def test_moving_avg_single_emission_fast_path(gpu, oracle_lib):
