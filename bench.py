@@ -192,8 +192,10 @@ def measure_streaming(fa, lib, chain, taps1, taps2, decim, fft_len,
 
 
 def measure_cpu_baseline(taps1, taps2, decim, fft_len, fixed_sample):
-    """oracle chain (the CPU restatement, kind 'port') on all host cores,
-    bounded to ~10 s of work."""
+    """Vectorized oracle chain (kind 'port') on all host cores, bounded to
+    ~10 s of work: the fused-tap algorithm (same as the GPU chain) over
+    deinterleaved planes, AVX2 FMA via omp simd — a defensible tuned-CPU
+    number, not the scalar restatement."""
     import oracle
     cores = os.cpu_count() or 1
     rng = np.random.default_rng(0x5D5D5D5D)
@@ -206,7 +208,7 @@ def measure_cpu_baseline(taps1, taps2, decim, fft_len, fixed_sample):
     t0 = time.perf_counter()
     while True:
         _, c = oracle.chain_cf32(taps1, taps2, decim, fft_len, x,
-                                 capture=False, nthreads=0)
+                                 capture=False, nthreads=0, fast=True)
         consumed += c
         passes += 1
         dt = time.perf_counter() - t0
@@ -219,7 +221,8 @@ def measure_cpu_baseline(taps1, taps2, decim, fft_len, fixed_sample):
         "cores": cores,
         "kind": "port",
         "sample": f"{passes} pass(es) over {n} Complex32 samples through "
-                  f"the oracle chain, OpenMP {cores} threads, {dt:.1f}s",
+                  f"the VECTORIZED oracle chain (fused 253-tap, AVX2 FMA "
+                  f"omp simd), OpenMP {cores} threads, {dt:.1f}s",
     }
 
 

@@ -103,6 +103,9 @@ def _load():
     lib.oracle_chain_cf32.restype = sz
     lib.oracle_chain_cf32.argtypes = [f32p, sz, f32p, sz, sz, sz,
                                       vp, sz, vp, sz, ctypes.c_int]
+    lib.oracle_chain_cf32_fast.restype = sz
+    lib.oracle_chain_cf32_fast.argtypes = [f32p, sz, f32p, sz, sz, sz,
+                                           vp, sz, vp, sz, ctypes.c_int]
     _lib = lib
     return lib
 
@@ -342,8 +345,11 @@ def pfb_channelizer(num_channels, decim, taps, inp, out_cap_per_chan):
     return out.reshape(num_channels, out_cap_per_chan)[:, :prod]
 
 
-def chain_cf32(taps1, taps2, decim, fft_len, inp, capture=True, nthreads=0):
-    """FIR(taps1) -> decim-FIR(taps2, decim) -> fft_len-pt forward DFT."""
+def chain_cf32(taps1, taps2, decim, fft_len, inp, capture=True, nthreads=0,
+               fast=False):
+    """FIR(taps1) -> decim-FIR(taps2, decim) -> fft_len-pt forward DFT.
+    fast=True uses the vectorized fused-tap variant (bench baseline leg
+    only — reassociated sums, tolerance-equal to the strict chain)."""
     lib = _load()
     taps1 = np.ascontiguousarray(taps1, np.float32)
     taps2 = np.ascontiguousarray(taps2, np.float32)
@@ -351,7 +357,8 @@ def chain_cf32(taps1, taps2, decim, fft_len, inp, capture=True, nthreads=0):
     y1 = inp.size + 1 - taps1.size
     frames = max(0, (y1 + 1 - taps2.size)) // decim // fft_len
     out = np.zeros(frames * fft_len, CF32) if capture else None
-    consumed = lib.oracle_chain_cf32(
+    fn = lib.oracle_chain_cf32_fast if fast else lib.oracle_chain_cf32
+    consumed = fn(
         _f32p(taps1), taps1.size, _f32p(taps2), taps2.size, decim, fft_len,
         _c(inp), inp.size,
         _c(out) if capture else None, out.size if capture else 0, nthreads)

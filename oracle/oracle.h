@@ -187,6 +187,17 @@ size_t oracle_chain_cf32(const float* taps1, size_t n_taps1,
                          ocf32* out_spectra, size_t n_out_cap,
                          int nthreads);
 
+/* Vectorized (AVX2 FMA via omp simd) fused-tap variant of the chain,
+ * used ONLY as bench.py's cpu_baseline leg — the defensible tuned-CPU
+ * number. Same fused algorithm as the GPU chain (taps convolved in f64);
+ * reassociated f32 sums, so tolerance-compare against the strict chain. */
+size_t oracle_chain_cf32_fast(const float* taps1, size_t n_taps1,
+                              const float* taps2, size_t n_taps2,
+                              size_t decim, size_t fft_len,
+                              const ocf32* in, size_t n_in,
+                              ocf32* out_spectra, size_t n_out_cap,
+                              int nthreads);
+
 #ifdef __cplusplus
 }
 #endif

@@ -242,6 +242,13 @@ def test_chain_matches_composition(oracle_lib):
     # within fp tolerance; per-frame computation is deterministic)
     ch4, _ = o.chain_cf32(t1, t2, 4, 1024, inp, nthreads=4)
     np.testing.assert_array_equal(ch, ch4)
+    # the vectorized baseline leg (fused taps, reassociated AVX2 sums)
+    # computes the same chain within fp tolerance — it is a real
+    # baseline, not a shortcut
+    chf, consf = o.chain_cf32(t1, t2, 4, 1024, inp, fast=True, nthreads=4)
+    assert consf == consumed and chf.size == ch.size
+    rel = np.linalg.norm(chf - ref) / np.linalg.norm(ref)
+    assert rel < 1e-4, rel
 
 
 def test_moving_avg_semantics(oracle_lib):
