@@ -235,6 +235,11 @@ def main():
 
     if not torch.cuda.is_available():
         raise SystemExit("bench.py needs a HIP device (no CPU fallback)")
+    ndev = torch.cuda.device_count()
+    if local >= ndev:  # oversubscribed test runs (2 ranks on a 1-GPU box)
+        log(f"rank {rank}: only {ndev} device(s); mapping local {local} "
+            f"-> {local % ndev}")
+        local = local % ndev
     torch.cuda.set_device(local)
 
     import futuresdr_amd as fa
