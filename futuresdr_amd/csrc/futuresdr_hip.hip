@@ -653,11 +653,17 @@ __global__ __launch_bounds__(MFIR32_BLOCK) void k_fir_mfma32_tpl(
 #define MDFIR_BLOCK 256
 #define MDFIR_TILE 1024 /* decimated outputs per block */
 
-/* forward declaration (defined with the FFT kernel below): in-block
- * 1024-pt forward Stockham FFT over swizzled LDS ping/pong buffers */
+/* forward declarations (defined with the FFT kernel below): in-block
+ * forward Stockham FFT over swizzled LDS ping/pong buffers. fft_pow2_fwd
+ * runs any pow2 n (radix-4 + radix-2 tail) with tpf threads striding the
+ * butterflies and returns the buffer holding the result; ALL threads of
+ * the block must call it with the same n (block-wide barriers inside). */
+__device__ float2* fft_pow2_fwd(float2* a, float2* b,
+                                const float2* __restrict__ twid, int n,
+                                int tf, int tpf);
 __device__ void fft1024_block(float2* ping, float2* pong,
                               const float2* __restrict__ twid, int tf);
-/* result lands in `pong` (5 stages = 5 swaps); read pong[fft_swz(i)] */
+/* fft1024_block result lands in `pong` (5 swaps); read pong[fft_swz(i)] */
 
 template <int KKD>
 __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
@@ -773,9 +779,10 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
     const float2* __restrict__ in, float2* __restrict__ out,
     const float* __restrict__ rtv /* [4][KKD], rtv[v][u] = rt[4u+v] */,
     long long n_out, long long n_in_valid,
-    const float2* __restrict__ twid /* 1024-entry forward table */,
-    float* __restrict__ mag_out /* nullable |X|^2 */) {
-    static_assert(MDFIR_TILE == 1024, "one tile == one 1024-pt FFT frame");
+    const float2* __restrict__ twid /* fft_len-entry forward table */,
+    float* __restrict__ mag_out /* nullable |X|^2 */,
+    int fft_len /* pow2 64..1024; tile = 1024/fft_len frames */) {
+    static_assert(MDFIR_TILE == 1024, "tile == 1024 decimated outputs");
     static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
     const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
     const unsigned SPm = (elemsP + 31u) & ~31u;
@@ -873,20 +880,31 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
         __syncthreads(); /* phase planes are dead; reuse them as FFT LDS */
         float2* ping = (float2*)planes;       /* 1024 float2 = 8 KB */
         float2* pong = ping + 1024;           /* fits in 4*SPm floats */
+        const unsigned Lm = (unsigned)fft_len - 1u;
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
-            int pos = wave * 256 + 16 * row + r16; /* y2 index in tile */
-            ping[fft_swz((unsigned)pos)] = make_float2(cre[q], cim[q]);
+            unsigned pos = wave * 256 + 16 * row + r16; /* y2 idx in tile */
+            ping[(pos & ~Lm) | fft_swz(pos & Lm)] =
+                make_float2(cre[q], cim[q]);
         }
         __syncthreads();
-        fft1024_block(ping, pong, twid, tid); /* result in pong */
-        for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
-            long long o = out_base + i;
-            if (o < n_out) {
-                float2 v = pong[fft_swz((unsigned)i)];
-                if (out) out[o] = v; /* null = NullSink'd spectra */
-                if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+        { /* one FFT per frame; frames share the block in lockstep */
+            const int F = 1024 / fft_len;
+            const int tpf = MDFIR_BLOCK / F;
+            const int fl = tid / tpf, tfr = tid - fl * tpf;
+            float2* res = fft_pow2_fwd(ping + (size_t)fl * fft_len,
+                                       pong + (size_t)fl * fft_len, twid,
+                                       fft_len, tfr, tpf) -
+                          (size_t)fl * fft_len; /* uniform base */
+            for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+                long long o = out_base + i;
+                if (o < n_out) {
+                    float2 v =
+                        res[((unsigned)i & ~Lm) | fft_swz((unsigned)i & Lm)];
+                    if (out) out[o] = v; /* null = NullSink'd spectra */
+                    if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+                }
             }
         }
         __syncthreads();
@@ -1562,17 +1580,14 @@ __device__ __forceinline__ float2 cmul_tw(float2 a, float2 w, int inverse) {
 /* In-block 1024-pt forward FFT (unnormalized), 256 threads, for the
  * chain's fused decim+FFT kernel. Same radix-4 DIF scheme and fft_swz
  * LDS swizzle as k_fft_stockham. */
-__device__ void fft1024_block(float2* ping, float2* pong,
-                              const float2* __restrict__ twid,
-                              int tf /* thread-in-frame, stride 256 */) {
-    const int n = 1024;
-    float2* a = ping;
-    float2* b = pong;
+__device__ float2* fft_pow2_fwd(float2* a, float2* b,
+                                const float2* __restrict__ twid, int n,
+                                int tf, int tpf) {
     int scur = 1;
     int ncur = n;
     while (ncur >= 4) {
         const int m4 = ncur >> 2;
-        for (int bf = tf; bf < n / 4; bf += 256) {
+        for (int bf = tf; bf < n / 4; bf += tpf) {
             const int p = bf / scur;
             const int q = bf - p * scur;
             const int tw = n / ncur;
@@ -1596,8 +1611,27 @@ __device__ void fft1024_block(float2* ping, float2* pong,
         ncur >>= 2;
         __syncthreads();
     }
-    /* 5 radix-4 stages = 5 ping/pong swaps: the result is in `pong`
-     * (callers read pong[fft_swz(i)]). */
+    if (ncur == 2) { /* radix-2 tail for odd log2n */
+        for (int bf = tf; bf < n / 2; bf += tpf) {
+            const int p = bf / scur;
+            const int q = bf - p * scur;
+            float2 xa = a[fft_swz(q + scur * p)];
+            float2 xb = a[fft_swz(q + scur * (p + 1))];
+            b[fft_swz(q + scur * 2 * p)] = f2_add(xa, xb);
+            b[fft_swz(q + scur * (2 * p + 1))] =
+                cmulf(f2_sub(xa, xb), twid[(size_t)p * (n / 2)]);
+        }
+        float2* t = a; a = b; b = t;
+        __syncthreads();
+    }
+    return a;
+}
+
+__device__ void fft1024_block(float2* ping, float2* pong,
+                              const float2* __restrict__ twid,
+                              int tf /* thread-in-frame, stride 256 */) {
+    (void)fft_pow2_fwd(ping, pong, twid, 1024, tf, 256);
+    /* 5 radix-4 stages = 5 ping/pong swaps: result in `pong`. */
 }
 
 __global__ __launch_bounds__(256) void k_fft_stockham(
@@ -2967,23 +3001,28 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
         out2 = c->d_null;
     }
     const char* ffz = getenv("FSDR_CHAIN_FFTFUSE");
-    if (c->fused && L == 1024 && c->fused->kk_mfma == 80 &&
+    const bool fft_fusable =
+        (L == 64 || L == 128 || L == 256 || L == 512 || L == 1024);
+    if (c->fused && fft_fusable && c->fused->kk_mfma &&
         !c->fft->inverse && !c->fft->fft_shift && c->fft->norm == 0.f &&
         (!ffz || atoi(ffz) != 0)) {
-        /* single kernel: fused 253-tap decimating filter + in-block
-         * 1024-pt FFT (+ optional |X|^2) — y2 never touches HBM */
+        /* single kernel: fused decimating filter + in-block per-frame
+         * FFT (+ optional |X|^2) — y2 never touches HBM. Tile = 1024
+         * decimated outputs = 1024/L frames. */
+        const int KK = c->fused->kk_mfma;
         long long tiles = ((long long)prod + MDFIR_TILE - 1) / MDFIR_TILE;
         long long cap = 8192; /* ~2 tiles/block at 2^26: best measured */
         if (const char* e = getenv("FSDR_FIR_GRID_CAP")) cap = atoll(e);
         int grid = (int)std::min<long long>(tiles, cap);
-        unsigned elemsP = MDFIR_TILE + 80 + 8;
-        size_t lds = (4 * (size_t)((elemsP + 31u) & ~31u) + 4 * (80 + 16)) *
-                     sizeof(float);
+        unsigned elemsP = MDFIR_TILE + KK + 8;
+        size_t lds = (4 * (size_t)((elemsP + 31u) & ~31u) +
+                      4 * ((size_t)KK + 16)) * sizeof(float);
         float2* spec_dst = (float2*)d_out; /* null + mag-only: skip the
                                               discarded spectra write */
         if (!spec_dst && !d_mag) spec_dst = out2;
         const char* b512 = getenv("FSDR_CHAIN_BLOCK512");
-        if (b512 && atoi(b512) != 0 && prod % 2048 == 0) {
+        if (b512 && atoi(b512) != 0 && L == 1024 && KK == 80 &&
+            prod % 2048 == 0) {
             long long tiles2 = (long long)prod / 2048;
             int grid2 = (int)std::min<long long>(tiles2, cap);
             unsigned eP2 = 2048 + 80 + 8;
@@ -2999,11 +3038,27 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
             HIP_TRY(hipGetLastError());
             return FSDR_OK;
         }
-        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_tpl<80>),
-                           dim3(grid), dim3(MDFIR_BLOCK), lds, st,
-                           (const float2*)d_in, spec_dst, c->fused->d_mtaps,
-                           (long long)prod, (long long)n_in,
-                           (const float2*)c->fft->d_twid, (float*)d_mag);
+#define CHAIN_TPL_CASE(KV)                                                \
+    case KV:                                                              \
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_tpl<KV>),    \
+                           dim3(grid), dim3(MDFIR_BLOCK), lds, st,        \
+                           (const float2*)d_in, spec_dst,                 \
+                           c->fused->d_mtaps, (long long)prod,            \
+                           (long long)n_in,                               \
+                           (const float2*)c->fft->d_twid, (float*)d_mag,  \
+                           (int)L);                                       \
+        break;
+        switch (KK) {
+            CHAIN_TPL_CASE(20)
+            CHAIN_TPL_CASE(32)
+            CHAIN_TPL_CASE(48)
+            CHAIN_TPL_CASE(80)
+            CHAIN_TPL_CASE(144)
+            default:
+                set_err("bad chain mfma K");
+                return FSDR_ERR_INVALID;
+        }
+#undef CHAIN_TPL_CASE
         HIP_TRY(hipGetLastError());
         return FSDR_OK;
     }

@@ -655,15 +655,17 @@ def test_wlan_sync_short_metric_chain(gpu, oracle_lib):
 
 # ---------------- randomized chain fuzz --------------------------------
 
-@pytest.mark.parametrize("seed", range(6))
+@pytest.mark.parametrize("seed", range(10))
 def test_chain_fuzz_vs_oracle(gpu, oracle_lib, seed):
     """Randomized tap counts / FFT lengths / stream sizes through the
-    fused chain vs the two-stage oracle (covers both the single-kernel
-    fft_len==1024 path and the split path)."""
+    fused chain vs the two-stage oracle. The single-kernel fused path now
+    engages for every pow2 fft_len 64..1024 and every MFMA K template
+    (20/32/48/80/144), so this fuzz covers the generalized kernel, not
+    just the 1024/K=80 bench shape."""
     r = rng(10_000 + seed)
     nt1 = int(r.integers(8, 250))
     nt2 = int(r.integers(8, 250))
-    fft_len = int(r.choice([256, 512, 1024]))
+    fft_len = int(r.choice([64, 128, 256, 512, 1024]))
     n_in = int(r.integers(4 * fft_len * 2 + nt1 + nt2, 200_000))
     t1 = r.uniform(-1, 1, nt1).astype(np.float32)
     t2 = r.uniform(-1, 1, nt2).astype(np.float32)
