@@ -109,6 +109,18 @@ def _load():
     lib.fsdr_divide_mag_dev.restype = ctypes.c_int
     lib.fsdr_divide_mag_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
                                         ctypes.POINTER(sz)]
+    lib.fsdr_wlan_rx_create.restype = vp
+    lib.fsdr_wlan_rx_create.argtypes = []
+    lib.fsdr_wlan_rx_destroy.argtypes = [vp]
+    lib.fsdr_wlan_sync_short_run.restype = sz
+    lib.fsdr_wlan_sync_short_run.argtypes = [vp, vp, vp, vp, sz, vp, sz,
+                                             vp, vp, sz,
+                                             ctypes.POINTER(sz),
+                                             ctypes.POINTER(sz)]
+    lib.fsdr_wlan_sync_long_run.restype = sz
+    lib.fsdr_wlan_sync_long_run.argtypes = [vp, vp, sz, vp, vp, sz, vp,
+                                            sz, vp, vp, sz,
+                                            ctypes.POINTER(sz)]
     lib.fsdr_wlan_moving_sum_dev.restype = ctypes.c_int
     lib.fsdr_wlan_moving_sum_dev.argtypes = [vp, sz, vp, sz, sz,
                                              ctypes.c_int, vp,
@@ -215,6 +227,10 @@ def _check(rc):
 def _f32(a):
     a = np.ascontiguousarray(a, np.float32)
     return a, a.ctypes.data_as(ctypes.POINTER(ctypes.c_float))
+
+
+def _c(a):
+    return ctypes.c_void_p(a.ctypes.data)
 
 
 def version():
@@ -609,6 +625,63 @@ def wlan_moving_sum_host(inp, length):
 
     out, p = _dev_roundtrip(call, [inp], inp.dtype, n_out)
     return out[:p]
+
+
+class WlanRx:
+    """WLAN rx front end host state machines (config 5): SyncShort
+    (sync_short.rs:92-150) + SyncLong (sync_long.rs:96-185, GPU 64-tap
+    correlator). Stateful like the reference blocks."""
+
+    def __init__(self):
+        self._h = _load().fsdr_wlan_rx_create()
+        if not self._h:
+            raise FsdrError("wlan rx create failed: "
+                            + _load().fsdr_last_error().decode())
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _load().fsdr_wlan_rx_destroy(self._h)
+            self._h = None
+
+    def sync_short(self, sig, abs48, cor, max_tags=64):
+        """Returns (frame_samples, tags) where tags is a list of
+        (output_index, coarse_freq_offset)."""
+        lib = _load()
+        sig = np.ascontiguousarray(sig, CF32)
+        abs48 = np.ascontiguousarray(abs48, CF32)
+        cor = np.ascontiguousarray(cor, np.float32)
+        n = min(sig.size, abs48.size, cor.size)
+        out = np.zeros(n, CF32)
+        tag_idx = np.zeros(max_tags, np.uintp)
+        tag_freq = np.zeros(max_tags, np.float32)
+        n_tags = ctypes.c_size_t()
+        consumed = ctypes.c_size_t()
+        prod = lib.fsdr_wlan_sync_short_run(
+            self._h, _c(sig), _c(abs48), _c(cor), n, _c(out), out.size,
+            _c(tag_idx), _c(tag_freq), max_tags, ctypes.byref(n_tags),
+            ctypes.byref(consumed))
+        tags = [(int(tag_idx[i]), float(tag_freq[i]))
+                for i in range(n_tags.value)]
+        return out[:prod], tags
+
+    def sync_long(self, frame_samples, tags, max_frames=16):
+        """Returns (symbol_stream, frames) where frames is a list of
+        (correlator_offset, fine_freq_offset)."""
+        lib = _load()
+        x = np.ascontiguousarray(frame_samples, CF32)
+        tag_idx = np.array([t[0] for t in tags], np.uintp)
+        tag_freq = np.array([t[1] for t in tags], np.float32)
+        out = np.zeros(x.size + 128 * max(1, len(tags)), CF32)
+        f_off = np.zeros(max_frames, np.uintp)
+        f_freq = np.zeros(max_frames, np.float32)
+        nf = ctypes.c_size_t()
+        prod = lib.fsdr_wlan_sync_long_run(
+            self._h, _c(x), x.size, _c(tag_idx), _c(tag_freq),
+            len(tags), _c(out), out.size, _c(f_off), _c(f_freq),
+            max_frames, ctypes.byref(nf))
+        frames = [(int(f_off[i]), float(f_freq[i]))
+                  for i in range(nf.value)]
+        return out[:prod], frames
 
 
 def cmul_host(a, b):

@@ -2914,6 +2914,264 @@ extern "C" int fsdr_rotator_dev(const void* d_in, void* d_out, size_t n,
     return FSDR_OK;
 }
 
+/* ================= WLAN rx front end (config 5) ======================= *
+ * Host-side state machines mirroring examples/wlan/src/sync_short.rs
+ * (:2-5 constants THRESHOLD 0.56 / MIN_GAP 480 / MAX_SAMPLES 540*80;
+ * :92-150 Search/Found/Copy loop) and sync_long.rs (:3 SEARCH_WINDOW
+ * 320; :18-50 Correlator::sync; :136-178 Sync/Copy). The sequential
+ * state machines run on the host exactly like the reference blocks; the
+ * SyncLong correlator's 320 64-tap complex dot products run on the GPU
+ * (k_fir_ccf32 through an internal filter handle). */
+
+/* the 802.11a long-training correlator taps — sync_long.rs:188-253 */
+static const float2 WLAN_LONG[64] = {
+    {1.3868f, -0.0000f},  {-0.0455f, 1.0679f},  {0.3528f, 0.9865f},
+    {0.8594f, -0.7348f},  {0.1874f, -0.2475f},  {0.5309f, 0.7784f},
+    {-1.0218f, 0.4897f},  {-0.3401f, 0.9423f},  {0.8657f, 0.2298f},
+    {0.4734f, -0.0362f},  {0.0088f, 1.0207f},   {-1.2142f, 0.4205f},
+    {0.2172f, 0.5195f},   {0.5207f, 0.1326f},   {-0.1995f, -1.4259f},
+    {1.0583f, 0.0363f},   {0.5547f, 0.5547f},   {0.3277f, -0.8728f},
+    {-0.5077f, -0.3488f}, {-1.1650f, -0.5789f}, {0.7297f, -0.8197f},
+    {0.6173f, -0.1253f},  {-0.5353f, -0.7214f}, {-0.5011f, 0.1935f},
+    {-0.3110f, 1.3392f},  {-1.0818f, 0.1470f},  {-1.1300f, 0.1820f},
+    {0.6663f, 0.6571f},   {-0.0249f, -0.4773f}, {-0.8155f, -1.0218f},
+    {0.8140f, -0.9396f},  {0.1090f, -0.8662f},  {-1.3868f, -0.0000f},
+    {0.1090f, 0.8662f},   {0.8140f, 0.9396f},   {-0.8155f, 1.0218f},
+    {-0.0249f, 0.4773f},  {0.6663f, -0.6571f},  {-1.1300f, -0.1820f},
+    {-1.0818f, -0.1470f}, {-0.3110f, -1.3392f}, {-0.5011f, -0.1935f},
+    {-0.5353f, 0.7214f},  {0.6173f, 0.1253f},   {0.7297f, 0.8197f},
+    {-1.1650f, 0.5789f},  {-0.5077f, 0.3488f},  {0.3277f, 0.8728f},
+    {0.5547f, -0.5547f},  {1.0583f, -0.0363f},  {-0.1995f, 1.4259f},
+    {0.5207f, -0.1326f},  {0.2172f, -0.5195f},  {-1.2142f, -0.4205f},
+    {0.0088f, -1.0207f},  {0.4734f, 0.0362f},   {0.8657f, -0.2298f},
+    {-0.3401f, -0.9423f}, {-1.0218f, -0.4897f}, {0.5309f, -0.7784f},
+    {0.1874f, 0.2475f},   {0.8594f, 0.7348f},   {0.3528f, -0.9865f},
+    {-0.0455f, -1.0679f},
+};
+
+#define WLAN_THRESHOLD 0.56f
+#define WLAN_MIN_GAP 480
+#define WLAN_MAX_SAMPLES (540 * 80)
+#define WLAN_SEARCH_WINDOW 320
+
+struct fsdr_wlan_rx {
+    /* SyncShort state (sync_short.rs:8-12,34) */
+    int ss_state = 0;      /* 0 Search, 1 Found, 2 Copy */
+    size_t ss_copied = 0;
+    float ss_foffset = 0.f;
+    bool ss_above = false;
+    bool ss_pending = false;
+    float ss_pending_freq = 0.f;
+    /* SyncLong GPU correlator */
+    fsdr_filter* corr = nullptr;
+    void* d_win = nullptr;
+    void* d_cor = nullptr;
+};
+
+extern "C" fsdr_wlan_rx* fsdr_wlan_rx_create(void) {
+    if (!have_gpu()) { set_err("no HIP device"); return nullptr; }
+    fsdr_wlan_rx* rx = new fsdr_wlan_rx();
+    /* FirCC computes y[i] = sum_t in[i+t]*taps[63-t]; the reference
+     * correlator is sum_k in[i+k]*LONG[k] (sync_long.rs:21-27), so the
+     * filter taps are LONG reversed. */
+    fsdr_cf32 rev[64];
+    for (int i = 0; i < 64; i++) {
+        rev[i].re = WLAN_LONG[63 - i].x;
+        rev[i].im = WLAN_LONG[63 - i].y;
+    }
+    rx->corr = fsdr_fir_ccf32_create(rev, 64);
+    if (!rx->corr ||
+        hipMalloc(&rx->d_win, (WLAN_SEARCH_WINDOW + 64) * sizeof(float2)) !=
+            hipSuccess ||
+        hipMalloc(&rx->d_cor, WLAN_SEARCH_WINDOW * sizeof(float2)) !=
+            hipSuccess) {
+        set_err("wlan rx alloc failed");
+        fsdr_filter_destroy(rx->corr);
+        if (rx->d_win) (void)hipFree(rx->d_win);
+        delete rx;
+        return nullptr;
+    }
+    return rx;
+}
+
+extern "C" void fsdr_wlan_rx_destroy(fsdr_wlan_rx* rx) {
+    if (!rx) return;
+    fsdr_filter_destroy(rx->corr);
+    if (rx->d_win) (void)hipFree(rx->d_win);
+    if (rx->d_cor) (void)hipFree(rx->d_cor);
+    delete rx;
+}
+
+/* SyncShort actor loop (sync_short.rs:92-150) over aligned spans of the
+ * delayed signal, the 48-sample complex autocorrelation average, and
+ * the correlation metric. Emits the frame-sample stream plus
+ * "wifi_start" tags (output index + coarse freq offset). Returns
+ * produced; *consumed_out = input samples consumed (== n unless out or
+ * tag capacity limited). */
+extern "C" size_t fsdr_wlan_sync_short_run(
+    fsdr_wlan_rx* rx, const fsdr_cf32* sig, const fsdr_cf32* abs48,
+    const float* cor, size_t n, fsdr_cf32* out, size_t out_cap,
+    size_t* tag_idx, float* tag_freq, size_t tag_cap, size_t* n_tags,
+    size_t* consumed_out) {
+    size_t i = 0, o = 0, nt = 0;
+    while (i < n && o < out_cap) {
+        switch (rx->ss_state) {
+            case 0: /* Search */
+                if (cor[i] > WLAN_THRESHOLD) rx->ss_state = 1;
+                break;
+            case 1: /* Found */
+                if (cor[i] > WLAN_THRESHOLD) {
+                    float fo = -atan2f(abs48[i].im, abs48[i].re) / 16.0f;
+                    rx->ss_state = 2;
+                    rx->ss_copied = 0;
+                    rx->ss_foffset = fo;
+                    rx->ss_above = false;
+                    rx->ss_pending = true;
+                    rx->ss_pending_freq = fo;
+                } else {
+                    rx->ss_state = 0;
+                }
+                break;
+            case 2: { /* Copy(n_copied, f_offset, last_above) */
+                if (cor[i] > WLAN_THRESHOLD) {
+                    if (rx->ss_above && rx->ss_copied > WLAN_MIN_GAP) {
+                        /* resync (sync_short.rs:110-117) */
+                        float fo =
+                            -atan2f(abs48[i].im, abs48[i].re) / 16.0f;
+                        rx->ss_copied = 0;
+                        rx->ss_foffset = fo;
+                        rx->ss_above = false;
+                        rx->ss_pending = true;
+                        rx->ss_pending_freq = fo;
+                        i++;
+                        continue;
+                    }
+                    rx->ss_above = true;
+                } else {
+                    rx->ss_above = false;
+                }
+                if (rx->ss_copied == 0 && rx->ss_pending) {
+                    if (nt < tag_cap) {
+                        tag_idx[nt] = o;
+                        tag_freq[nt] = rx->ss_pending_freq;
+                        nt++;
+                    }
+                    rx->ss_pending = false;
+                }
+                float ang = rx->ss_foffset * (float)rx->ss_copied;
+                float s, c;
+                __builtin_sincosf(ang, &s, &c);
+                out[o].re = sig[i].re * c - sig[i].im * s;
+                out[o].im = sig[i].re * s + sig[i].im * c;
+                o++;
+                if (rx->ss_copied + 1 == WLAN_MAX_SAMPLES)
+                    rx->ss_state = 0;
+                else
+                    rx->ss_copied++;
+                break;
+            }
+        }
+        i++;
+    }
+    if (n_tags) *n_tags = nt;
+    if (consumed_out) *consumed_out = i;
+    return o;
+}
+
+/* SyncLong over a tagged span (sync_long.rs:96-185): for each
+ * "wifi_start" tag, run the GPU 64-tap correlator over the
+ * SEARCH_WINDOW, pick the top-2 |cor|^2 peaks (stable order, pair
+ * sorted by index — :38-48), copy 128 samples rotated by the fine freq
+ * offset, then strip the 16-sample CP from each 80-sample symbol until
+ * the next tag (leftover < 80 dropped, like the reference's m<80
+ * consume). Needs a GPU. Returns symbols*64 (+128/frame) samples
+ * written; frame_off[f] = the correlator offset chosen for frame f. */
+extern "C" size_t fsdr_wlan_sync_long_run(
+    fsdr_wlan_rx* rx, const fsdr_cf32* in, size_t n,
+    const size_t* tag_idx, const float* tag_freq, size_t num_tags,
+    fsdr_cf32* out, size_t out_cap, size_t* frame_off, float* frame_freq,
+    size_t frame_cap, size_t* num_frames) {
+    (void)tag_freq;
+    size_t o = 0, nf = 0;
+    for (size_t t = 0; t < num_tags; t++) {
+        size_t T = tag_idx[t];
+        size_t T2 = (t + 1 < num_tags) ? tag_idx[t + 1] : n;
+        if (T2 - T < WLAN_SEARCH_WINDOW + 128) continue; /* :143 */
+        /* Correlator::sync on in[T .. T+SEARCH_WINDOW+63] (GPU) */
+        if (hipMemcpy(rx->d_win, in + T,
+                      (WLAN_SEARCH_WINDOW + 63) * sizeof(float2),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_err("wlan h2d failed");
+            break;
+        }
+        fsdr_filter_result r;
+        if (fsdr_filter_dev(rx->corr, rx->d_win, WLAN_SEARCH_WINDOW + 63,
+                            rx->d_cor, WLAN_SEARCH_WINDOW, nullptr,
+                            &r) != FSDR_OK)
+            break;
+        (void)hipStreamSynchronize(nullptr);
+        float2 corv[WLAN_SEARCH_WINDOW];
+        if (hipMemcpy(corv, rx->d_cor,
+                      WLAN_SEARCH_WINDOW * sizeof(float2),
+                      hipMemcpyDeviceToHost) != hipSuccess)
+            break;
+        /* top-2 by |cor|^2, stable (first occurrence wins ties), pair
+         * ordered by index (:38-44) */
+        int i0 = 0, i1 = -1;
+        float m0 = -1.f, m1 = -1.f;
+        for (int i = 0; i < WLAN_SEARCH_WINDOW; i++) {
+            float m = corv[i].x * corv[i].x + corv[i].y * corv[i].y;
+            if (m > m0) {
+                m1 = m0; i1 = i0;
+                m0 = m; i0 = i;
+            } else if (m > m1) {
+                m1 = m; i1 = i;
+            }
+        }
+        int first = i0 < i1 ? i0 : i1;
+        int second = i0 < i1 ? i1 : i0;
+        /* freq = arg(cor[first] * conj(cor[second])) / 64  (:46-48) */
+        float2 a = corv[first], b = corv[second];
+        float pr = a.x * b.x + a.y * b.y;
+        float pi = a.y * b.x - a.x * b.y;
+        float freq = atan2f(pi, pr) / 64.0f;
+        size_t off = (size_t)first;
+        if (o + 128 > out_cap) break;
+        for (int i = 0; i < 128; i++) { /* :147-151 */
+            float s, c;
+            __builtin_sincosf((float)i * freq, &s, &c);
+            fsdr_cf32 x = in[T + off + i];
+            out[o + i].re = x.re * c - x.im * s;
+            out[o + i].im = x.re * s + x.im * c;
+        }
+        o += 128;
+        if (nf < frame_cap) {
+            if (frame_off) frame_off[nf] = off;
+            if (frame_freq) frame_freq[nf] = freq;
+            nf++;
+        }
+        /* Copy state (:162-177): 80-sample symbols -> 64 samples each */
+        size_t cur = T + off + 128;
+        size_t n_copied = 0;
+        while (cur + 80 <= T2 && o + 64 <= out_cap) {
+            for (int k = 0; k < 64; k++) {
+                float ang =
+                    (float)(n_copied * 80 + 128 + 16 + (size_t)k) * freq;
+                float s, c;
+                __builtin_sincosf(ang, &s, &c);
+                fsdr_cf32 x = in[cur + 16 + k];
+                out[o + k].re = x.re * c - x.im * s;
+                out[o + k].im = x.re * s + x.im * c;
+            }
+            o += 64;
+            cur += 80;
+            n_copied++;
+        }
+    }
+    if (num_frames) *num_frames = nf;
+    return o;
+}
+
 /* ================= chain ============================================== */
 
 struct fsdr_chain {

@@ -171,6 +171,31 @@ int fsdr_wlan_moving_sum_dev(const void* d_in, size_t n_in, void* d_out,
                              size_t n_out, size_t len, int is_complex,
                              void* stream, size_t* produced);
 
+/* WLAN rx front end (config 5): the host-side SyncShort state machine
+ * (examples/wlan/src/sync_short.rs:92-150; THRESHOLD 0.56, MIN_GAP 480,
+ * MAX_SAMPLES 540*80) and SyncLong (sync_long.rs:96-185; SEARCH_WINDOW
+ * 320, 64-tap LONG correlator run on the GPU, top-2 peak sync, CP
+ * strip). sync_short consumes aligned spans of (delayed signal, 48-avg
+ * autocorrelation, correlation metric) and emits frame samples +
+ * "wifi_start" tags; sync_long consumes that tagged stream and emits
+ * 128 preamble samples + 64-sample OFDM symbols per frame, ready for
+ * the 64-pt Fft block (rx.rs:84-103). */
+typedef struct fsdr_wlan_rx fsdr_wlan_rx;
+fsdr_wlan_rx* fsdr_wlan_rx_create(void);
+void fsdr_wlan_rx_destroy(fsdr_wlan_rx* rx);
+size_t fsdr_wlan_sync_short_run(fsdr_wlan_rx* rx, const fsdr_cf32* sig,
+                                const fsdr_cf32* abs48, const float* cor,
+                                size_t n, fsdr_cf32* out, size_t out_cap,
+                                size_t* tag_idx, float* tag_freq,
+                                size_t tag_cap, size_t* n_tags,
+                                size_t* consumed_out);
+size_t fsdr_wlan_sync_long_run(fsdr_wlan_rx* rx, const fsdr_cf32* in,
+                               size_t n, const size_t* tag_idx,
+                               const float* tag_freq, size_t num_tags,
+                               fsdr_cf32* out, size_t out_cap,
+                               size_t* frame_off, float* frame_freq,
+                               size_t frame_cap, size_t* num_frames);
+
 /* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
 int fsdr_dev_alloc(void** d_ptr, size_t bytes);
 int fsdr_dev_free(void* d_ptr);
