@@ -1562,6 +1562,59 @@ __global__ void k_resamp_cf32(const float2* __restrict__ in,
     }
 }
 
+/* LDS-tiled resampler (polyphase_resampling_fir.rs:108-118 semantics,
+ * identical output math to k_resamp_cf32): a tile of RS_TILE outputs
+ * stages its whole input span into padded SoA LDS planes once, then
+ * each lane runs its tpp-tap phase dot product from LDS — removes the
+ * per-output global gather of the naive kernel. Fallback to
+ * k_resamp_cf32 when the span exceeds the LDS budget (host decides). */
+#define RS_BLOCK 256
+#define RS_R 2
+#define RS_TILE (RS_BLOCK * RS_R)
+
+__global__ __launch_bounds__(RS_BLOCK) void k_resamp_tiled_cf32(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ taps, int n_taps, int interp, int decim,
+    long long n_out, long long n_in_valid, int span) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* s_re = (float*)smem;
+    float* s_im = s_re + plane_floats((unsigned)span);
+    float* s_taps = s_im + plane_floats((unsigned)span);
+    for (int i = threadIdx.x; i < n_taps; i += RS_BLOCK)
+        s_taps[i] = taps[i];
+    const int tpp = n_taps / interp;
+    const long long tiles = (n_out + RS_TILE - 1) / RS_TILE;
+    for (long long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
+        const long long out_base = tile * RS_TILE;
+        const long long in_base = out_base * decim / interp;
+        __syncthreads(); /* previous tile's reads done before restage */
+        for (int i = threadIdx.x; i < span; i += RS_BLOCK) {
+            long long g = in_base + i;
+            float2 v = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+            s_re[lds_pad((unsigned)i)] = v.x;
+            s_im[lds_pad((unsigned)i)] = v.y;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int r = 0; r < RS_R; r++) {
+            long long k = out_base + threadIdx.x + r * RS_BLOCK;
+            if (k >= n_out) break;
+            int bank = (int)((k * decim) % interp);
+            unsigned rel = (unsigned)(k * decim / interp - in_base);
+            float sre = 0.f, sim = 0.f;
+            const float* tb = s_taps + bank;
+#pragma unroll 4
+            for (int t = 0; t < tpp; t++) {
+                unsigned a = lds_pad(rel + (unsigned)t);
+                float h = tb[interp * (tpp - t - 1)];
+                sre = fmaf(s_re[a], h, sre);
+                sim = fmaf(s_im[a], h, sim);
+            }
+            out[k] = make_float2(sre, sim);
+        }
+    }
+}
+
 /* ================= FFT: radix-4 Stockham in LDS ======================= *
  * Unnormalized DFT, rustfft convention (forward e^{-2pi i kn/N}) — the
  * reference Fft block's math (src/blocks/fft.rs:190-194). Stockham
@@ -2503,6 +2556,26 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
         case K_RESAMP_CF32: {
             *r = resamp_status(f->interp, f->decim, f->n_taps, n_in, n_out);
             if (r->produced == 0) return FSDR_OK;
+            size_t tpp = f->n_taps / f->interp;
+            long long span = (long long)(RS_TILE - 1) * f->decim /
+                                 f->interp + tpp + 2;
+            size_t lds_t = (2 * (size_t)plane_floats((unsigned)span) +
+                            f->n_taps) * sizeof(float);
+            const char* rt = getenv("FSDR_RESAMP_TILED");
+            if (lds_t <= 64 * 1024 && (!rt || atoi(rt) != 0)) {
+                long long tiles =
+                    ((long long)r->produced + RS_TILE - 1) / RS_TILE;
+                int grid = (int)std::min<long long>(tiles, 256 * 64);
+                hipLaunchKernelGGL(k_resamp_tiled_cf32, dim3(grid),
+                                   dim3(RS_BLOCK), lds_t, st,
+                                   (const float2*)d_in, (float2*)d_out,
+                                   f->d_taps, (int)f->n_taps,
+                                   (int)f->interp, (int)f->decim,
+                                   (long long)r->produced, (long long)n_in,
+                                   (int)span);
+                HIP_TRY(hipGetLastError());
+                return FSDR_OK;
+            }
             size_t lds = f->n_taps * sizeof(float);
             hipLaunchKernelGGL(k_resamp_cf32,
                                dim3(grid_for((long long)r->produced, 256)),
