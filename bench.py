@@ -63,6 +63,8 @@ def parse_args():
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--skip-streaming", action="store_true",
                    help="skip the PCIe-fed ring streaming leg")
+    p.add_argument("--skip-config3", action="store_true",
+                   help="skip the config-3 (FM resampler chain) leg")
     p.add_argument("--streaming-chunk", type=int, default=1 << 22,
                    help="ring chunk size in samples for the streaming leg")
     p.add_argument("--streaming-chunks", type=int, default=64,
@@ -189,6 +191,80 @@ def measure_streaming(fa, lib, chain, taps1, taps2, decim, fft_len,
     finally:
         lib.fsdr_dev_free(d_mag)
         lib.fsdr_ring_destroy(ring)
+
+
+def measure_config3(fa, lib, torch):
+    """BASELINE configs[2]: HipFir(127-tap lowpass) -> PolyphaseResampler
+    4:1 -> 1024-pt FFT on a resident 2^26-sample batch. Reports MS/s of
+    chain input plus the resampler kernel's own roofline block."""
+    beta = fa.kaiser_beta(1e-4)
+    taps = fa.lowpass_kaiser_n(127, beta, 0.1)
+    rtaps = fa.lowpass_kaiser_n(128, beta, 0.2)  # n_taps % interp == 0
+    fir = fa.Fir(taps)
+    rs = fa.Resampler(1, 4, rtaps)
+    fft = fa.Fft(1024)
+    S = 1 << 26
+    y1 = S + 1 - taps.size
+    # resamp_status: prod = ((y1+1-128)*1 - 1)//4 floored to interp mult
+    y2 = (y1 + 1 - rtaps.size - 1) // 4
+    frames = y2 // 1024
+    st = torch.cuda.current_stream()
+    d_in = alloc_dev(lib, S * 8)
+    d_y1 = alloc_dev(lib, y1 * 8)
+    d_y2 = alloc_dev(lib, y2 * 8)
+    d_sp = alloc_dev(lib, frames * 1024 * 8)
+    try:
+        fa.fill_uniform_dev(d_in.value, S, seed=7, stream=st.cuda_stream)
+
+        def step():
+            fir.filter_dev(d_in.value, S, d_y1.value, y1,
+                           stream=st.cuda_stream)
+            rs.filter_dev(d_y1.value, y1, d_y2.value, y2,
+                          stream=st.cuda_stream)
+            fft.bulk_dev(d_y2.value, d_sp.value, frames,
+                         stream=st.cuda_stream)
+        for _ in range(3):
+            step()
+        torch.cuda.synchronize()
+        reps = 10
+        ev0 = torch.cuda.Event(enable_timing=True)
+        ev1 = torch.cuda.Event(enable_timing=True)
+        ev0.record(st)
+        for _ in range(reps):
+            step()
+        ev1.record(st)
+        torch.cuda.synchronize()
+        ms = ev0.elapsed_time(ev1) / reps
+        # resampler kernel alone (its roofline)
+        ev0.record(st)
+        for _ in range(reps):
+            rs.filter_dev(d_y1.value, y1, d_y2.value, y2,
+                          stream=st.cuda_stream)
+        ev1.record(st)
+        torch.cuda.synchronize()
+        rs_ms = ev0.elapsed_time(ev1) / reps
+        rs_prod = (y2 // 1) * 1
+        rs_tf = rs_prod * rtaps.size * 4 / (rs_ms * 1e-3) / 1e12
+        return {
+            "value": round(S * 1e3 / ms / 1e6, 2),
+            "unit": "MSample/s",
+            "workload": "config3 fm: fir127 -> resamp 4:1 (128 taps) -> "
+                        "fft1024, resident 2^26 batch",
+            "ms_per_step": round(ms, 4),
+            "resampler_roofline": {
+                "bound": "mfma",
+                "achieved": round(rs_tf, 2),
+                "peak": FP32_PEAK_TFLOPS,
+                "unit": "TFLOP/s",
+                "frac": round(rs_tf / FP32_PEAK_TFLOPS, 4),
+                "traffic": None,
+                "kernel": "k_resamp_tiled_cf32",
+                "ms_per_launch": round(rs_ms, 4),
+            },
+        }
+    finally:
+        for p in (d_in, d_y1, d_y2, d_sp):
+            lib.fsdr_dev_free(p)
 
 
 def measure_cpu_baseline(taps1, taps2, decim, fft_len, fixed_sample):
@@ -343,6 +419,10 @@ def main():
                                           args.decim, args.fft,
                                           args.streaming_chunk,
                                           args.streaming_chunks)
+        config3 = None
+        if n_gpus == 1 and not args.skip_config3:
+            log("measuring config-3 (FM resampler chain) leg...")
+            config3 = measure_config3(fa, lib, torch)
         result = {
             "metric": "MSample/s through 127-tap C32 FIR→decim4→"
                       "1k-FFT flowgraph @1/2/4/8 GPU",
@@ -376,6 +456,7 @@ def main():
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
             "streaming": streaming,
+            "config3": config3,
         }
         print(json.dumps(result), flush=True)
 

@@ -102,6 +102,8 @@ def _load():
     lib.fsdr_filter_host.argtypes = [vp, vp, sz, vp, sz, rp]
     lib.fsdr_filter_dev.restype = ctypes.c_int
     lib.fsdr_filter_dev.argtypes = [vp, vp, sz, vp, sz, vp, rp]
+    lib.fsdr_fft_bulk_dev.restype = ctypes.c_int
+    lib.fsdr_fft_bulk_dev.argtypes = [vp, vp, vp, vp, sz, vp]
     lib.fsdr_filter_destroy.argtypes = [vp]
     lib.fsdr_cmul_conj_dev.restype = ctypes.c_int
     lib.fsdr_cmul_conj_dev.argtypes = [vp, sz, vp, sz, vp, sz, vp,
@@ -369,6 +371,13 @@ class Fft(Filter):
             np_ = ctypes.pointer(ctypes.c_float(normalize))
         super().__init__(_load().fsdr_fft_cf32_create(
             length, int(inverse), int(fft_shift), np_))
+
+    def bulk_dev(self, d_in, d_out, frames, d_mag=0, stream=None):
+        """One launch over `frames` device-resident frames (the batch
+        path; fsdr_filter_dev keeps the reference's 32-frame quantum)."""
+        _check(_load().fsdr_fft_bulk_dev(
+            self._h, ctypes.c_void_p(d_in), ctypes.c_void_p(d_out),
+            ctypes.c_void_p(d_mag), frames, ctypes.c_void_p(stream or 0)))
 
 
 class Mag2(Filter):

@@ -2524,6 +2524,22 @@ static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
     return FSDR_OK;
 }
 
+/* Bulk batch FFT: the GPU-native many-frames path (what the chain uses
+ * internally). The 32-frames-per-work cap in fsdr_filter_dev mirrors the
+ * reference block's work() quantum (fft.rs:56); a batch resident in HBM
+ * has no reason to launch 512 times. d_mag optional fused |X|^2. */
+extern "C" int fsdr_fft_bulk_dev(fsdr_filter* f, const void* d_in,
+                                 void* d_out, void* d_mag, size_t frames,
+                                 void* stream) {
+    REQUIRE_GPU();
+    if (!f || f->kind != K_FFT_CF32) {
+        set_err("fft_bulk: not an fft filter");
+        return FSDR_ERR_INVALID;
+    }
+    return launch_fft(f, d_in, d_out, frames, (hipStream_t)stream,
+                      (float*)d_mag);
+}
+
 extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                                void* d_out, size_t n_out, void* stream,
                                fsdr_filter_result* r) {

@@ -128,6 +128,12 @@ int fsdr_filter_host(fsdr_filter* f, const void* in, size_t n_in,
 int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
                     void* d_out, size_t n_out, void* stream,
                     fsdr_filter_result* r);
+/* Bulk batch FFT over `frames` fft_len-sized device-resident frames in
+ * one launch (the 32-frame cap in fsdr_filter_dev mirrors the reference
+ * work() quantum, fft.rs:56 — this is the GPU-native batch path the
+ * chain uses). d_mag: optional fused |X|^2 output (nullable). */
+int fsdr_fft_bulk_dev(fsdr_filter* f, const void* d_in, void* d_out,
+                      void* d_mag, size_t frames, void* stream);
 void fsdr_filter_destroy(fsdr_filter* f);
 
 /* Combine (2-input zip-map), complex-multiply variant. m = produced. */
