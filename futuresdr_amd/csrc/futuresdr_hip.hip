@@ -528,23 +528,44 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
             load_tile(tile + gridDim.x);
 
         const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-        v4f cre = {0.f, 0.f, 0.f, 0.f};
-        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        /* two interleaved accumulator pairs: a single C chain issues one
+         * dependent MFMA per result latency (~2x the 8-clk issue rate),
+         * stalling the SIMD half the time — the round-1 WAIT_ANY ~50%.
+         * Even/odd K-steps accumulate independently and merge at the end
+         * (fp32 reassociation, within the parity tolerance). */
+        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int s = 0; s < KK / 4; s++) {
+        for (int s = 0; s < KK / 4 - 1; s += 2) {
+            float a_re0 = s_re[mfma_swz(ab + 4 * s)];
+            float a_im0 = s_im[mfma_swz(ab + 4 * s)];
+            float a_re1 = s_re[mfma_swz(ab + 4 * (s + 1))];
+            float a_im1 = s_im[mfma_swz(ab + 4 * (s + 1))];
+            cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re0, bfrag[s],
+                                                        cre0, 0, 0, 0);
+            cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im0, bfrag[s],
+                                                        cim0, 0, 0, 0);
+            cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re1, bfrag[s + 1],
+                                                        cre1, 0, 0, 0);
+            cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im1, bfrag[s + 1],
+                                                        cim1, 0, 0, 0);
+        }
+        if (KK / 4 & 1) {
+            const int s = KK / 4 - 1;
             float a_re = s_re[mfma_swz(ab + 4 * s)];
             float a_im = s_im[mfma_swz(ab + 4 * s)];
-            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
-                                                       0, 0, 0);
-            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
-                                                       0, 0, 0);
+            cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s],
+                                                        cre0, 0, 0, 0);
+            cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s],
+                                                        cim0, 0, 0, 0);
         }
         /* C layout: col = lane&15, row = (lane>>4)*4 + q (cdna4 16x16) */
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
             long long o = out_base + (long long)wave * 256 + 16 * row + r16;
-            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+            if (o < n_out)
+                out[o] = make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
         }
         __syncthreads();
     }
@@ -722,7 +743,13 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
         }
     };
     const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
+    /* Two interleaved accumulator pairs per half: a single C chain can
+     * only issue a dependent MFMA every result latency (~2x the issue
+     * rate), stalling the SIMD half the time — the round-1 WAIT_ANY
+     * ~50%. Even/odd K-steps accumulate independently per phase and
+     * merge at the output (fp32 reassociation, within tolerance). */
+    auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
+                         v4f& cim1) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
             const float* pre = planes + (unsigned)vloc * SPm;
@@ -736,13 +763,28 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
                                               FFT/staging phases of
                                               co-resident blocks (T5) */
 #pragma unroll
-            for (int s = 0; s < KKD / 4; s++) {
+            for (int s = 0; s < KKD / 4 - 1; s += 2) {
+                float a_re0 = pre[mfma_swz(ab + 4 * s)];
+                float a_im0 = pim[mfma_swz(ab + 4 * s)];
+                float a_re1 = pre[mfma_swz(ab + 4 * (s + 1))];
+                float a_im1 = pim[mfma_swz(ab + 4 * (s + 1))];
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re0, bfrag[s], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im0, bfrag[s], cim0, 0, 0, 0);
+                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re1, bfrag[s + 1], cre1, 0, 0, 0);
+                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im1, bfrag[s + 1], cim1, 0, 0, 0);
+            }
+            if (KKD / 4 & 1) {
+                const int s = KKD / 4 - 1;
                 float a_re = pre[mfma_swz(ab + 4 * s)];
                 float a_im = pim[mfma_swz(ab + 4 * s)];
-                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[s], cre, 0, 0, 0);
-                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[s], cim, 0, 0, 0);
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim0, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
         }
@@ -752,23 +794,24 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * MDFIR_TILE;
-        v4f cre = {0.f, 0.f, 0.f, 0.f};
-        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
         load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
-        mfma_half(0, cre, cim);
+        mfma_half(0, cre0, cim0, cre1, cim1);
         __syncthreads();
         write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
-        mfma_half(1, cre, cim);
+        mfma_half(1, cre0, cim0, cre1, cim1);
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
             long long o = out_base + (long long)wave * 256 + 16 * row + r16;
-            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+            if (o < n_out)
+                out[o] = make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
         }
         __syncthreads();
     }
@@ -835,7 +878,13 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
         }
     };
     const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
+    /* Two interleaved accumulator pairs per half: a single C chain can
+     * only issue a dependent MFMA every result latency (~2x the issue
+     * rate), stalling the SIMD half the time — the round-1 WAIT_ANY
+     * ~50%. Even/odd K-steps accumulate independently per phase and
+     * merge at the output (fp32 reassociation, within tolerance). */
+    auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
+                         v4f& cim1) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
             const float* pre = planes + (unsigned)vloc * SPm;
@@ -849,13 +898,28 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
                                               FFT/staging phases of
                                               co-resident blocks (T5) */
 #pragma unroll
-            for (int s = 0; s < KKD / 4; s++) {
+            for (int s = 0; s < KKD / 4 - 1; s += 2) {
+                float a_re0 = pre[mfma_swz(ab + 4 * s)];
+                float a_im0 = pim[mfma_swz(ab + 4 * s)];
+                float a_re1 = pre[mfma_swz(ab + 4 * (s + 1))];
+                float a_im1 = pim[mfma_swz(ab + 4 * (s + 1))];
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re0, bfrag[s], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im0, bfrag[s], cim0, 0, 0, 0);
+                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re1, bfrag[s + 1], cre1, 0, 0, 0);
+                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im1, bfrag[s + 1], cim1, 0, 0, 0);
+            }
+            if (KKD / 4 & 1) {
+                const int s = KKD / 4 - 1;
                 float a_re = pre[mfma_swz(ab + 4 * s)];
                 float a_im = pim[mfma_swz(ab + 4 * s)];
-                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[s], cre, 0, 0, 0);
-                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[s], cim, 0, 0, 0);
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim0, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
         }
@@ -865,18 +929,18 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * MDFIR_TILE;
-        v4f cre = {0.f, 0.f, 0.f, 0.f};
-        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
         load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
-        mfma_half(0, cre, cim);
+        mfma_half(0, cre0, cim0, cre1, cim1);
         __syncthreads();
         write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
-        mfma_half(1, cre, cim);
+        mfma_half(1, cre0, cim0, cre1, cim1);
         __syncthreads(); /* phase planes are dead; reuse them as FFT LDS */
         float2* ping = (float2*)planes;       /* 1024 float2 = 8 KB */
         float2* pong = ping + 1024;           /* fits in 4*SPm floats */
@@ -886,7 +950,7 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
             int row = k4 * 4 + q;
             unsigned pos = wave * 256 + 16 * row + r16; /* y2 idx in tile */
             ping[(pos & ~Lm) | fft_swz(pos & Lm)] =
-                make_float2(cre[q], cim[q]);
+                make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
         }
         __syncthreads();
         { /* one FFT per frame; frames share the block in lockstep */
@@ -971,7 +1035,13 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
         }
     };
     const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
+    /* Two interleaved accumulator pairs per half: a single C chain can
+     * only issue a dependent MFMA every result latency (~2x the issue
+     * rate), stalling the SIMD half the time — the round-1 WAIT_ANY
+     * ~50%. Even/odd K-steps accumulate independently per phase and
+     * merge at the output (fp32 reassociation, within tolerance). */
+    auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
+                         v4f& cim1) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
             const float* pre = planes + (unsigned)vloc * SPm;
@@ -985,13 +1055,28 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
                                               FFT/staging phases of
                                               co-resident blocks (T5) */
 #pragma unroll
-            for (int s = 0; s < KKD / 4; s++) {
+            for (int s = 0; s < KKD / 4 - 1; s += 2) {
+                float a_re0 = pre[mfma_swz(ab + 4 * s)];
+                float a_im0 = pim[mfma_swz(ab + 4 * s)];
+                float a_re1 = pre[mfma_swz(ab + 4 * (s + 1))];
+                float a_im1 = pim[mfma_swz(ab + 4 * (s + 1))];
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re0, bfrag[s], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im0, bfrag[s], cim0, 0, 0, 0);
+                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re1, bfrag[s + 1], cre1, 0, 0, 0);
+                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im1, bfrag[s + 1], cim1, 0, 0, 0);
+            }
+            if (KKD / 4 & 1) {
+                const int s = KKD / 4 - 1;
                 float a_re = pre[mfma_swz(ab + 4 * s)];
                 float a_im = pim[mfma_swz(ab + 4 * s)];
-                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[s], cre, 0, 0, 0);
-                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[s], cim, 0, 0, 0);
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim0, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
         }
@@ -1001,18 +1086,18 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)2048 < n_out; tile += gridDim.x) {
         const long long out_base = tile * 2048;
-        v4f cre = {0.f, 0.f, 0.f, 0.f};
-        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
         load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
-        mfma_half(0, cre, cim);
+        mfma_half(0, cre0, cim0, cre1, cim1);
         __syncthreads();
         write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)2048 < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
-        mfma_half(1, cre, cim);
+        mfma_half(1, cre0, cim0, cre1, cim1);
         __syncthreads(); /* phase planes are dead; reuse them as FFT LDS:
                             frame f ping = fbase + f*2048, pong = +1024 */
         float2* fbase = (float2*)planes; /* 4096 float2 = 32 KB */
@@ -1022,7 +1107,7 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
             int pos = wave * 256 + 16 * row + r16; /* y2 index in tile */
             int fr = pos >> 10, idx = pos & 1023;
             fbase[fr * 2048 + fft_swz((unsigned)idx)] =
-                make_float2(cre[q], cim[q]);
+                make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
         }
         __syncthreads();
         {
