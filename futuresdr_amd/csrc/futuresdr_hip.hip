@@ -2232,6 +2232,12 @@ extern "C" fsdr_filter* fsdr_resamp_cf32_create(size_t interp, size_t decim,
         return nullptr;
     }
     f->n_taps_padded = (int)n_taps;
+    /* interp==1, decim==4 (the config-3 "4:1" shape): y[k] =
+     * sum_t x[4k+t]*h[T-1-t] is the decimating FIR evaluated with its
+     * window origin shifted by D-1, so route it to the MFMA decim
+     * kernel with the input pointer rebased (launch site). */
+    if (interp == 1 && decim == 4 && n_taps <= 512)
+        f->sub = fsdr_decim_fir_cf32_create(4, taps, n_taps);
     return f;
 }
 
@@ -2657,6 +2663,15 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
         case K_RESAMP_CF32: {
             *r = resamp_status(f->interp, f->decim, f->n_taps, n_in, n_out);
             if (r->produced == 0) return FSDR_OK;
+            const char* dmf = getenv("FSDR_DECIM_MFMA");
+            if (f->sub && f->sub->kk_mfma && (!dmf || atoi(dmf) != 0)) {
+                /* 1:4 shape on the MFMA decim kernel: shift the window
+                 * origin back by D-1 (that kernel never reads rel < 3,
+                 * so the rebased pointer stays in bounds) */
+                return launch_decim_cf32(f->sub,
+                                         (const float2*)d_in - 3, d_out,
+                                         r->produced, n_in + 3, st);
+            }
             size_t tpp = f->n_taps / f->interp;
             long long span = (long long)(RS_TILE - 1) * f->decim /
                                  f->interp + tpp + 2;
