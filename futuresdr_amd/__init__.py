@@ -88,6 +88,10 @@ def _load():
     lib.fsdr_moving_avg_create.restype = vp
     lib.fsdr_pfb_channelizer_create.restype = vp
     lib.fsdr_pfb_channelizer_create.argtypes = [sz, f32p, sz, ctypes.c_float]
+    lib.fsdr_pfb_channelizer_stream_dev.restype = ctypes.c_int
+    lib.fsdr_pfb_channelizer_stream_dev.argtypes = [vp, vp, sz, vp, sz,
+                                                    vp, ctypes.POINTER(sz),
+                                                    ctypes.POINTER(sz)]
     lib.fsdr_pfb_channelizer_run_dev.restype = ctypes.c_int
     lib.fsdr_pfb_channelizer_run_dev.argtypes = [vp, vp, sz, vp, sz, vp,
                                                  ctypes.POINTER(sz)]
@@ -201,6 +205,19 @@ def _load():
     lib.fsdr_ring_reader_release_consumed.restype = ctypes.c_int
     lib.fsdr_ring_reader_release_consumed.argtypes = [vp, sz, vp]
     lib.fsdr_ring_destroy.argtypes = [vp]
+    lib.fsdr_ring_d2h_create.restype = vp
+    lib.fsdr_ring_d2h_create.argtypes = [sz, sz, sz]
+    lib.fsdr_ring_d2h_writer_acquire.restype = ctypes.c_int
+    lib.fsdr_ring_d2h_writer_acquire.argtypes = [vp, ctypes.POINTER(vp),
+                                                 ctypes.POINTER(sz), vp]
+    lib.fsdr_ring_d2h_writer_commit.restype = ctypes.c_int
+    lib.fsdr_ring_d2h_writer_commit.argtypes = [vp, sz, vp]
+    lib.fsdr_ring_d2h_reader_acquire.restype = ctypes.c_int
+    lib.fsdr_ring_d2h_reader_acquire.argtypes = [vp, ctypes.POINTER(vp),
+                                                 ctypes.POINTER(sz)]
+    lib.fsdr_ring_d2h_reader_release.restype = ctypes.c_int
+    lib.fsdr_ring_d2h_reader_release.argtypes = [vp]
+    lib.fsdr_ring_d2h_destroy.argtypes = [vp]
     _lib = lib
     return lib
 
@@ -397,21 +414,54 @@ class XlatingFir(Filter):
 
 
 class PfbChannelizer(Filter):
-    """PFB channelizer — pfb/channelizer.rs (maximally decimated)."""
+    """PFB channelizer — pfb/channelizer.rs (any oversample N/i;
+    decimation D = N/oversample). Stateful streaming via stream()."""
 
     def __init__(self, num_channels, taps, oversample_rate=1.0):
         self._taps_keep, p = _f32(taps)
         self.n = num_channels
+        self.decim = int(num_channels / oversample_rate)
         super().__init__(_load().fsdr_pfb_channelizer_create(
             num_channels, p, self._taps_keep.size, oversample_rate))
+
+    def stream(self, chunk):
+        """Feed one host chunk through the stateful streaming path;
+        returns ([num_channels, produced] array, consumed). Unconsumed
+        samples must be re-presented by the caller (slab semantics)."""
+        lib = _load()
+        chunk = np.ascontiguousarray(chunk, CF32)
+        cap = max(1, (chunk.size + self.n * self._tpf()) // self.decim)
+        d_in = ctypes.c_void_p()
+        d_out = ctypes.c_void_p()
+        _check(lib.fsdr_dev_alloc(ctypes.byref(d_in), chunk.size * 8))
+        _check(lib.fsdr_dev_alloc(ctypes.byref(d_out), self.n * cap * 8))
+        try:
+            _check(lib.fsdr_memcpy_h2d(
+                d_in, ctypes.c_void_p(chunk.ctypes.data), chunk.size * 8))
+            prod = ctypes.c_size_t()
+            cons = ctypes.c_size_t()
+            _check(lib.fsdr_pfb_channelizer_stream_dev(
+                self._h, d_in, chunk.size, d_out, cap, None,
+                ctypes.byref(prod), ctypes.byref(cons)))
+            _check(lib.fsdr_synchronize())
+            out = np.zeros(self.n * cap, CF32)
+            _check(lib.fsdr_memcpy_d2h(ctypes.c_void_p(out.ctypes.data),
+                                       d_out, self.n * cap * 8))
+            return (out.reshape(self.n, cap)[:, :prod.value], cons.value)
+        finally:
+            lib.fsdr_dev_free(d_in)
+            lib.fsdr_dev_free(d_out)
+
+    def _tpf(self):
+        return -(-self._taps_keep.size // self.n)
 
     def run(self, inp):
         """Bulk one-shot from zero state over a host span; returns an
         array of shape [num_channels, produced]."""
         lib = _load()
         inp = np.ascontiguousarray(inp, CF32)
-        tpf = -(-self._taps_keep.size // self.n)
-        cap = max(1, (inp.size - self.n * tpf) // self.n)
+        tpf = self._tpf()
+        cap = max(1, (inp.size - self.n * tpf) // self.decim)
         d_in = ctypes.c_void_p()
         d_out = ctypes.c_void_p()
         _check(lib.fsdr_dev_alloc(ctypes.byref(d_in), inp.size * 8))

@@ -1486,28 +1486,42 @@ __global__ void k_moving_sum(const float* __restrict__ in,
     }
 }
 
-/* ================= PFB channelizer (maximally decimated) ============== *
- * src/blocks/pfb/channelizer.rs (liquid-dsp scheme), bulk form for
- * oversample_rate = 1 from zero state: with D == N the round-robin
- * window algebra collapses to
- *   fft_buf[k][b] = sum_j x[N*(tpf+k-j) + N-1-b] * part[b][j]
- * (part = partition_filter_taps, utilities.rs:5-25; prefill consumes
- * N*tpf samples), followed by an unnormalized inverse FFT of size N per
- * step and a transpose to channel-major outputs. */
+/* ================= PFB channelizer ==================================== *
+ * src/blocks/pfb/channelizer.rs (liquid-dsp scheme), bulk closed form
+ * of the round-robin window state for ANY decimation D = N/oversample
+ * (channelizer.rs:126-210). Derivation: push p of the input stream
+ * lands in window w(p) = (N-1-p) mod N (decrement_base_index); after
+ * P = prefill + (k+1)*D pushes (prefill = N*tpf), output step k reads
+ * window b's newest-j element x[pmax(b) - j*N] with
+ *   pmax(b) = P-1 - ((P-1 - (N-1-b)) mod N)
+ * through filter i(b) = (b - base - 1) mod N, base = (N-1-P) mod N:
+ *   fft_buf[k][b] = sum_j x[pmax(b) - j*N] * part[i(b)][j]
+ * (part = partition_filter_taps, utilities.rs:5-25; the FirFilter's
+ * reversed taps turn the oldest->newest window into newest-j order).
+ * `p_before` = pushes before this launch's first step (global),
+ * `bufbase` = global stream index of in[0] (streaming carries the last
+ * N*tpf consumed samples in front). D == N collapses to the maximally
+ * decimated form (base stays N-1, i == b). */
 __global__ void k_pfb_dots(const float2* __restrict__ in,
                            float2* __restrict__ fb,
                            const float* __restrict__ part /* [N][tpf] */,
-                           int N, int tpf, long long steps) {
+                           int N, int tpf, int D, long long steps,
+                           long long p_before, long long bufbase) {
     long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
     long long total = steps * N;
     long long stride = (long long)gridDim.x * blockDim.x;
     for (; id < total; id += stride) {
         long long k = id / N;
         int b = (int)(id - k * N);
+        long long P = p_before + (k + 1) * (long long)D;
+        int base = (int)(((N - 1 - P) % N + N) % N);
+        int i = ((b - base - 1) % N + N) % N;
+        long long pmax =
+            P - 1 - (((P - 1 - (N - 1 - b)) % N + N) % N);
         float sre = 0.f, sim = 0.f;
         for (int j = 0; j < tpf; j++) {
-            float2 x = in[(long long)N * (tpf + k - j) + (N - 1 - b)];
-            float tap = part[b * tpf + j];
+            float2 x = in[pmax - (long long)j * N - bufbase];
+            float tap = part[i * tpf + j];
             sre = fmaf(x.x, tap, sre);
             sim = fmaf(x.y, tap, sim);
         }
@@ -2007,11 +2021,17 @@ struct fsdr_filter {
     void* d_out = nullptr;
     size_t d_in_bytes = 0, d_out_bytes = 0;
     size_t item_in = 8, item_out = 8;
-    /* dedicated kernel scratch (MovingAvg chunk partials). MUST be
-     * distinct from d_in: on the fsdr_filter_host path d_in holds the
-     * staged input while the partials kernel is still reading it. */
+    /* dedicated kernel scratch (MovingAvg chunk partials, PFB work
+     * buffer). MUST be distinct from d_in: on the fsdr_filter_host path
+     * d_in holds the staged input while the partials kernel is still
+     * reading it. */
     void* d_scratch = nullptr;
     size_t d_scratch_bytes = 0;
+    /* PFB streaming state: the last N*tpf CONSUMED samples (the window
+     * contents) + total pushes (channelizer.rs round-robin state) */
+    void* d_hist = nullptr;
+    size_t d_hist_bytes = 0;
+    size_t pfb_pushes = 0;
 };
 
 static int ensure_dev(void** p, size_t* cur, size_t want) {
@@ -2328,9 +2348,10 @@ extern "C" fsdr_filter* fsdr_pfb_channelizer_create(size_t num_channels,
         set_err("pfb: num_channels > 2 and taps.len() >= num_channels");
         return nullptr;
     }
-    if (oversample_rate != 1.0f) {
-        set_err("pfb: only oversample_rate == 1 (maximally decimated) is "
-                "implemented on the GPU path");
+    /* channelizer.rs:100-104: oversample_rate must be N/i, i in [1,N] */
+    if (!(oversample_rate > 0.f) ||
+        fmodf((float)num_channels, oversample_rate) != 0.f) {
+        set_err("pfb: oversample rate must be N/i for i in [1, N]");
         return nullptr;
     }
     if ((num_channels & (num_channels - 1)) != 0 || num_channels < 4 ||
@@ -2342,7 +2363,8 @@ extern "C" fsdr_filter* fsdr_pfb_channelizer_create(size_t num_channels,
     fsdr_filter* f = create_common(K_PFB);
     if (!f) return nullptr;
     f->width = num_channels;
-    f->decim = num_channels; /* D == N at oversample 1 */
+    /* channelizer.rs:105: decimation_factor = N / oversample_rate */
+    f->decim = (size_t)((float)num_channels / oversample_rate);
     size_t tpf = (n_taps + num_channels - 1) / num_channels;
     f->history = tpf;
     f->n_taps = n_taps;
@@ -2418,6 +2440,7 @@ extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (f->d_in) (void)hipFree(f->d_in);
     if (f->d_out) (void)hipFree(f->d_out);
     if (f->d_scratch) (void)hipFree(f->d_scratch);
+    if (f->d_hist) (void)hipFree(f->d_hist);
     delete f;
 }
 
@@ -2850,9 +2873,9 @@ extern "C" int fsdr_pfb_channelizer_run_dev(fsdr_filter* f,
         return FSDR_ERR_INVALID;
     }
     hipStream_t st = (hipStream_t)stream;
-    size_t N = f->width, tpf = f->history;
+    size_t N = f->width, tpf = f->history, D = f->decim;
     size_t prefill = N * tpf;
-    size_t steps = n_in > prefill ? (n_in - prefill) / N : 0;
+    size_t steps = n_in > prefill ? (n_in - prefill) / D : 0;
     if (steps > out_cap_per_chan) steps = out_cap_per_chan;
     if (produced_per_chan) *produced_per_chan = steps;
     if (steps == 0) return FSDR_OK;
@@ -2864,7 +2887,8 @@ extern "C" int fsdr_pfb_channelizer_run_dev(fsdr_filter* f,
                        dim3(grid_for((long long)(steps * N), 256)),
                        dim3(256), 0, st, (const float2*)d_in,
                        (float2*)f->d_in, (const float*)f->d_taps, (int)N,
-                       (int)tpf, (long long)steps);
+                       (int)tpf, (int)D, (long long)steps,
+                       (long long)prefill, (long long)0);
     HIP_TRY(hipGetLastError());
     rc = launch_fft(f->sub, f->d_in, f->d_out, steps, st);
     if (rc) return rc;
@@ -2874,6 +2898,79 @@ extern "C" int fsdr_pfb_channelizer_run_dev(fsdr_filter* f,
                        (float2*)d_out, (int)N, (long long)steps,
                        (long long)out_cap_per_chan);
     HIP_TRY(hipGetLastError());
+    return FSDR_OK;
+}
+
+/* Streaming channelizer: stateful like the reference block — carries
+ * the round-robin window contents (the last N*tpf CONSUMED samples) and
+ * the push count across calls, consuming in D-sample quanta after the
+ * N*tpf prefill (channelizer.rs:156-212). The unconsumed chunk
+ * remainder stays with the caller (pair with fsdr_ring's
+ * release_consumed carry). All work is enqueued on `stream`; use one
+ * stream per filter. */
+extern "C" int fsdr_pfb_channelizer_stream_dev(
+    fsdr_filter* f, const void* d_chunk, size_t n, void* d_out,
+    size_t out_cap_per_chan, void* stream, size_t* produced_per_chan,
+    size_t* consumed) {
+    REQUIRE_GPU();
+    if (!f || f->kind != K_PFB) {
+        set_err("not a pfb channelizer");
+        return FSDR_ERR_INVALID;
+    }
+    hipStream_t st = (hipStream_t)stream;
+    size_t N = f->width, tpf = f->history, D = f->decim;
+    size_t prefill = N * tpf, W = N * tpf;
+    size_t p = f->pfb_pushes;
+    size_t T = p < W ? p : W;
+    size_t fill = p < prefill ? std::min(prefill - p, n) : 0;
+    size_t steps = (p + n >= prefill) ? (p + n - prefill) / D -
+                                            (p > prefill ? (p - prefill) / D
+                                                         : 0)
+                                      : 0;
+    if (steps > out_cap_per_chan) steps = out_cap_per_chan;
+    size_t cons = fill + steps * D;
+    if (cons > n) cons = n; /* fill-only calls */
+    if (produced_per_chan) *produced_per_chan = steps;
+    if (consumed) *consumed = cons;
+    if (cons == 0) return FSDR_OK;
+    /* work buffer = [hist(T), chunk(cons)] */
+    int rc = ensure_dev(&f->d_scratch, &f->d_scratch_bytes,
+                        (T + cons) * 8);
+    if (rc) return rc;
+    if (T)
+        HIP_TRY(hipMemcpyAsync(f->d_scratch, f->d_hist, T * 8,
+                               hipMemcpyDeviceToDevice, st));
+    HIP_TRY(hipMemcpyAsync((char*)f->d_scratch + T * 8, d_chunk, cons * 8,
+                           hipMemcpyDeviceToDevice, st));
+    if (steps) {
+        rc = ensure_dev(&f->d_in, &f->d_in_bytes, steps * N * 8);
+        if (rc) return rc;
+        rc = ensure_dev(&f->d_out, &f->d_out_bytes, steps * N * 8);
+        if (rc) return rc;
+        hipLaunchKernelGGL(k_pfb_dots,
+                           dim3(grid_for((long long)(steps * N), 256)),
+                           dim3(256), 0, st, (const float2*)f->d_scratch,
+                           (float2*)f->d_in, (const float*)f->d_taps,
+                           (int)N, (int)tpf, (int)D, (long long)steps,
+                           (long long)(p + fill), (long long)(p - T));
+        HIP_TRY(hipGetLastError());
+        rc = launch_fft(f->sub, f->d_in, f->d_out, steps, st);
+        if (rc) return rc;
+        hipLaunchKernelGGL(k_pfb_scatter,
+                           dim3(grid_for((long long)(steps * N), 256)),
+                           dim3(256), 0, st, (const float2*)f->d_out,
+                           (float2*)d_out, (int)N, (long long)steps,
+                           (long long)out_cap_per_chan);
+        HIP_TRY(hipGetLastError());
+    }
+    /* new hist = last min(W, T+cons) samples of the work buffer */
+    size_t newT = std::min(W, T + cons);
+    rc = ensure_dev(&f->d_hist, &f->d_hist_bytes, W * 8);
+    if (rc) return rc;
+    HIP_TRY(hipMemcpyAsync(f->d_hist,
+                           (char*)f->d_scratch + (T + cons - newT) * 8,
+                           newT * 8, hipMemcpyDeviceToDevice, st));
+    f->pfb_pushes = p + cons;
     return FSDR_OK;
 }
 
@@ -3702,6 +3799,138 @@ extern "C" int fsdr_ring_reader_release(fsdr_ring* r) {
 }
 
 extern "C" void fsdr_ring_destroy(fsdr_ring* r) {
+    if (!r) return;
+    for (auto& b : r->bufs) {
+        if (b.host) (void)hipHostFree(b.host);
+        if (b.dev) (void)hipFree(b.dev);
+        if (b.ev) (void)hipEventDestroy(b.ev);
+        if (b.free_ev) (void)hipEventDestroy(b.free_ev);
+    }
+    if (r->copy_stream) (void)hipStreamDestroy(r->copy_stream);
+    delete r;
+}
+
+/* ================= D2H return ring ==================================== *
+ * The reverse direction of fsdr_ring: a device-side producer fills ring
+ * buffers, async D2H on the ring's copy stream hands them to a pinned
+ * host consumer — the vulkan d2h.rs Writer::submit / host Reader pair
+ * (d2h.rs:66,247-268) with the circuit's empty-buffer recirculation
+ * (d2h.rs:284-299). Writer: acquire(stream) (returns a device buffer;
+ * the caller's stream is made to wait for that buffer's previous D2H)
+ * -> fill on the stream -> commit(items, stream) (enqueues D2H after
+ * the producer's work). Reader: acquire (waits for the copy; yields the
+ * pinned host span) -> release (recycles). SPSC. */
+
+struct fsdr_ring_d2h {
+    size_t n_buffers, items_per_buffer, item_bytes;
+    std::vector<RingBuf> bufs; /* free_ev = producer-done event */
+    std::deque<int> empty_q, full_q;
+    int writer_cur = -1, reader_cur = -1;
+    std::mutex mu;
+    std::condition_variable cv;
+    hipStream_t copy_stream = nullptr;
+};
+
+extern "C" fsdr_ring_d2h* fsdr_ring_d2h_create(size_t n_buffers,
+                                               size_t items_per_buffer,
+                                               size_t item_bytes) {
+    if (!have_gpu()) { set_err("no HIP device"); return nullptr; }
+    if (n_buffers < 2 || items_per_buffer == 0 || item_bytes == 0) {
+        set_err("invalid ring parameters");
+        return nullptr;
+    }
+    fsdr_ring_d2h* r = new fsdr_ring_d2h();
+    r->n_buffers = n_buffers;
+    r->items_per_buffer = items_per_buffer;
+    r->item_bytes = item_bytes;
+    if (hipStreamCreate(&r->copy_stream) != hipSuccess) {
+        set_err("stream create failed");
+        delete r;
+        return nullptr;
+    }
+    size_t bytes = items_per_buffer * item_bytes;
+    for (size_t i = 0; i < n_buffers; i++) {
+        RingBuf b;
+        if (hipHostMalloc(&b.host, bytes) != hipSuccess ||
+            hipMalloc(&b.dev, bytes) != hipSuccess ||
+            hipEventCreate(&b.ev) != hipSuccess ||
+            hipEventCreate(&b.free_ev) != hipSuccess) {
+            set_err("ring buffer alloc failed");
+            r->bufs.push_back(b);
+            fsdr_ring_d2h_destroy(r);
+            return nullptr;
+        }
+        r->bufs.push_back(b);
+        r->empty_q.push_back((int)i);
+    }
+    return r;
+}
+
+extern "C" int fsdr_ring_d2h_writer_acquire(fsdr_ring_d2h* r,
+                                            void** dev_ptr, size_t* items,
+                                            void* stream) {
+    if (!r) return FSDR_ERR_INVALID;
+    std::unique_lock<std::mutex> lk(r->mu);
+    r->cv.wait(lk, [&] { return !r->empty_q.empty(); });
+    r->writer_cur = r->empty_q.front();
+    r->empty_q.pop_front();
+    lk.unlock();
+    RingBuf& b = r->bufs[r->writer_cur];
+    /* the producer must not overwrite dev before its previous D2H is
+     * drained (no-op for a never-recorded event) */
+    HIP_TRY(hipStreamWaitEvent((hipStream_t)stream, b.ev, 0));
+    *dev_ptr = b.dev;
+    *items = r->items_per_buffer;
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_ring_d2h_writer_commit(fsdr_ring_d2h* r, size_t items,
+                                           void* stream) {
+    if (!r || r->writer_cur < 0) return FSDR_ERR_INVALID;
+    RingBuf& b = r->bufs[r->writer_cur];
+    b.items = items;
+    HIP_TRY(hipEventRecord(b.free_ev, (hipStream_t)stream));
+    HIP_TRY(hipStreamWaitEvent(r->copy_stream, b.free_ev, 0));
+    HIP_TRY(hipMemcpyAsync(b.host, b.dev, items * r->item_bytes,
+                           hipMemcpyDeviceToHost, r->copy_stream));
+    HIP_TRY(hipEventRecord(b.ev, r->copy_stream));
+    {
+        std::lock_guard<std::mutex> lk(r->mu);
+        r->full_q.push_back(r->writer_cur);
+        r->writer_cur = -1;
+    }
+    r->cv.notify_all();
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_ring_d2h_reader_acquire(fsdr_ring_d2h* r,
+                                            void** host_ptr,
+                                            size_t* items) {
+    if (!r) return FSDR_ERR_INVALID;
+    std::unique_lock<std::mutex> lk(r->mu);
+    r->cv.wait(lk, [&] { return !r->full_q.empty(); });
+    r->reader_cur = r->full_q.front();
+    r->full_q.pop_front();
+    lk.unlock();
+    RingBuf& b = r->bufs[r->reader_cur];
+    HIP_TRY(hipEventSynchronize(b.ev));
+    *host_ptr = b.host;
+    *items = b.items;
+    return FSDR_OK;
+}
+
+extern "C" int fsdr_ring_d2h_reader_release(fsdr_ring_d2h* r) {
+    if (!r || r->reader_cur < 0) return FSDR_ERR_INVALID;
+    {
+        std::lock_guard<std::mutex> lk(r->mu);
+        r->empty_q.push_back(r->reader_cur);
+        r->reader_cur = -1;
+    }
+    r->cv.notify_all();
+    return FSDR_OK;
+}
+
+extern "C" void fsdr_ring_d2h_destroy(fsdr_ring_d2h* r) {
     if (!r) return;
     for (auto& b : r->bufs) {
         if (b.host) (void)hipHostFree(b.host);

@@ -152,9 +152,13 @@ int fsdr_rotator_dev(const void* d_in, void* d_out, size_t n,
 
 /* PfbChannelizer (src/blocks/pfb/channelizer.rs, liquid-dsp scheme):
  * splits a Complex32 stream into num_channels frequency channels.
- * GPU path implements the maximally-decimated case (oversample_rate = 1)
- * in bulk form from zero state; num_channels must be a power of two in
- * [4,4096]. Output is channel-major: out[c*out_cap_per_chan + k]. */
+ * Supports any oversample_rate = N/i (decimation D = N/oversample,
+ * channelizer.rs:100-105); num_channels must be a power of two in
+ * [4,4096] (FFT kernel). Output is channel-major:
+ * out[c*out_cap_per_chan + k]. run_dev = bulk from zero state;
+ * stream_dev carries the round-robin window state across calls and
+ * consumes in D quanta after the N*tpf prefill (pair with the ring's
+ * release_consumed carry for exact arbitrary-chunk streaming). */
 fsdr_filter* fsdr_pfb_channelizer_create(size_t num_channels,
                                          const float* taps, size_t n_taps,
                                          float oversample_rate);
@@ -162,6 +166,11 @@ int fsdr_pfb_channelizer_run_dev(fsdr_filter* f, const void* d_in,
                                  size_t n_in, void* d_out,
                                  size_t out_cap_per_chan, void* stream,
                                  size_t* produced_per_chan);
+int fsdr_pfb_channelizer_stream_dev(fsdr_filter* f, const void* d_chunk,
+                                    size_t n, void* d_out,
+                                    size_t out_cap_per_chan, void* stream,
+                                    size_t* produced_per_chan,
+                                    size_t* consumed);
 
 /* WLAN sync-short autocorrelation helpers (examples/wlan/src/bin/
  * rx.rs:73-96): a*conj(b) Combine and the sliding-SUM MovingAverage
@@ -270,6 +279,24 @@ int  fsdr_ring_reader_release_consumed(fsdr_ring* r, size_t consumed,
                                        void* stream);
 int  fsdr_ring_reader_release(fsdr_ring* r);
 void fsdr_ring_destroy(fsdr_ring* r);
+
+/* D2H return ring (vulkan d2h.rs Writer::submit / host Reader + the
+ * circuit's empty recirculation): device producer fills ring buffers on
+ * its stream, async D2H on the ring's copy stream hands pinned host
+ * spans to the consumer. Writer acquire takes the producer's stream so
+ * buffer reuse is ordered after the previous D2H without host syncs. */
+typedef struct fsdr_ring_d2h fsdr_ring_d2h;
+fsdr_ring_d2h* fsdr_ring_d2h_create(size_t n_buffers,
+                                    size_t items_per_buffer,
+                                    size_t item_bytes);
+int  fsdr_ring_d2h_writer_acquire(fsdr_ring_d2h* r, void** dev_ptr,
+                                  size_t* items, void* stream);
+int  fsdr_ring_d2h_writer_commit(fsdr_ring_d2h* r, size_t items,
+                                 void* stream);
+int  fsdr_ring_d2h_reader_acquire(fsdr_ring_d2h* r, void** host_ptr,
+                                  size_t* items);
+int  fsdr_ring_d2h_reader_release(fsdr_ring_d2h* r);
+void fsdr_ring_d2h_destroy(fsdr_ring_d2h* r);
 
 /* ---- Flowgraph driver (native C++ harness) ---------------------------- *
  * Minimal mirror of Flowgraph::add / Flowgraph::stream / Runtime::run for

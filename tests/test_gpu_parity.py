@@ -608,6 +608,53 @@ def test_pfb_channelizer_tone_isolation_gpu(gpu, oracle_lib):
     assert e[3] / e.sum() > 0.95
 
 
+@pytest.mark.parametrize("oversample", [2.0, 4.0])
+def test_pfb_channelizer_oversample_parity(gpu, oracle_lib, oversample):
+    """Oversampled (D = N/oversample < N): the round-robin base rotates
+    every step (channelizer.rs:126-135) — GPU bulk closed form vs the
+    oracle restatement with the same D."""
+    N = 8
+    D = int(N / oversample)
+    r = rng(163 + D)
+    taps = r.uniform(-1, 1, 64).astype(np.float32)
+    x = cplx(r, 4096)
+    got = gpu.PfbChannelizer(N, taps, oversample_rate=oversample).run(x)
+    ref = oracle_lib.pfb_channelizer(N, D, taps, x, got.shape[1] + 8)
+    assert got.shape[1] == ref.shape[1]
+    assert_close(got, ref, 1e-5)
+
+
+def test_pfb_channelizer_streaming(gpu, oracle_lib):
+    """Arbitrary chunk sizes through the stateful streaming path; the
+    caller re-presents the unconsumed tail (slab semantics). The
+    concatenated channel outputs equal the one-shot oracle."""
+    N, over = 8, 2.0
+    D = int(N / over)
+    r = rng(167)
+    taps = r.uniform(-1, 1, 64).astype(np.float32)
+    n_total = 6000
+    x = cplx(r, n_total)
+    ch = gpu.PfbChannelizer(N, taps, oversample_rate=over)
+    outs = []
+    leftover = np.zeros(0, np.complex64)
+    off = 0
+    while off < n_total or leftover.size:
+        take = int(r.integers(7, 700))
+        buf = np.concatenate([leftover, x[off:off + take]])
+        off += take
+        out, cons = ch.stream(buf)
+        assert cons <= buf.size
+        if out.shape[1]:
+            outs.append(out)
+        leftover = buf[cons:]
+        if off >= n_total and cons == 0:
+            break
+    got = np.concatenate(outs, axis=1)
+    ref = oracle_lib.pfb_channelizer(N, D, taps, x, got.shape[1] + 8)
+    assert got.shape[1] <= ref.shape[1]
+    assert_close(got, ref[:, :got.shape[1]], 1e-5)
+
+
 # ---------------- WLAN sync-short autocorrelation chain ----------------
 
 def test_wlan_ops_parity(gpu, oracle_lib):
@@ -763,6 +810,69 @@ def test_ring_chain_streaming_tiny_chunks(gpu, oracle_lib):
     r = rng(157)
     sizes = r.integers(40, 900, size=30).tolist()
     _chain_stream_case(gpu, oracle_lib, 64, sizes, 159)
+
+
+def test_ring_full_loop_h2d_chain_d2h(gpu, oracle_lib):
+    """The complete streaming loop: host chunks -> pinned H2D ring ->
+    fused chain on the compute stream -> D2H return ring -> host
+    consumer. Results equal the one-shot oracle chain. Exercises the
+    d2h ring's event ordering (producer-stream wait, copy-stream D2H)."""
+    lib = gpu.lib()
+    r = rng(171)
+    t1 = r.uniform(-1, 1, 127).astype(np.float32)
+    t2 = r.uniform(-1, 1, 127).astype(np.float32)
+    fft_len = 256
+    chunk = 9000
+    n_total = chunk * 6
+    x = cplx(r, n_total)
+    g_len = t1.size + t2.size - 1
+    reserved = g_len - 1 + 4 * fft_len + 4
+    out_cap = (reserved + chunk) // 4 + fft_len
+    chain = gpu.Chain(t1, t2, 4, fft_len)
+    ring = lib.fsdr_ring_create(4, chunk, 8, reserved)
+    d2h = lib.fsdr_ring_d2h_create(4, out_cap, 8)
+    assert ring and d2h
+    outs = []
+    try:
+        for off in range(0, n_total, chunk):
+            hp = ctypes.c_void_p()
+            items = ctypes.c_size_t()
+            assert lib.fsdr_ring_writer_acquire(
+                ring, ctypes.byref(hp), ctypes.byref(items)) == 0
+            ctypes.memmove(hp, ctypes.c_void_p(x[off:off + chunk]
+                                               .ctypes.data), chunk * 8)
+            assert lib.fsdr_ring_writer_commit(ring, chunk) == 0
+            dp = ctypes.c_void_p()
+            got_items = ctypes.c_size_t()
+            assert lib.fsdr_ring_reader_acquire(
+                ring, ctypes.byref(dp), ctypes.byref(got_items)) == 0
+            op = ctypes.c_void_p()
+            ocap = ctypes.c_size_t()
+            assert lib.fsdr_ring_d2h_writer_acquire(
+                d2h, ctypes.byref(op), ctypes.byref(ocap), None) == 0
+            cons, prod = chain.run_dev(dp.value, got_items.value,
+                                       op.value, ocap.value)
+            assert lib.fsdr_ring_d2h_writer_commit(d2h, prod, None) == 0
+            assert lib.fsdr_ring_reader_release_consumed(
+                ring, cons, None) == 0
+            hp2 = ctypes.c_void_p()
+            n2 = ctypes.c_size_t()
+            assert lib.fsdr_ring_d2h_reader_acquire(
+                d2h, ctypes.byref(hp2), ctypes.byref(n2)) == 0
+            if n2.value:
+                h = np.zeros(n2.value, np.complex64)
+                ctypes.memmove(ctypes.c_void_p(h.ctypes.data), hp2,
+                               n2.value * 8)
+                outs.append(h)
+            assert lib.fsdr_ring_d2h_reader_release(d2h) == 0
+        got = np.concatenate(outs)
+        ref, _ = oracle_lib.chain_cf32(t1, t2, 4, fft_len, x)
+        assert got.size == ref.size
+        rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+        assert rel < 2e-4, rel
+    finally:
+        lib.fsdr_ring_destroy(ring)
+        lib.fsdr_ring_d2h_destroy(d2h)
 
 
 def test_moving_avg_single_emission_fast_path(gpu, oracle_lib):

@@ -293,6 +293,21 @@ def test_pfb_channelizer_tone_isolation(oracle_lib):
         assert e[c0] / e.sum() > 0.95, (c0, e / e.sum())
 
 
+def test_pfb_channelizer_oversampled_tone_isolation(oracle_lib):
+    """Oversampled (D = N/2): output rate doubles but a tone at channel
+    c's center still lands in channel c (the round-robin state of
+    channelizer.rs:126-210 at D < N)."""
+    o = oracle_lib
+    N = 8
+    taps = o.kaiser_multirate_f32(N, 1, 4, 1e-3)
+    m = np.arange(4096, dtype=np.float64)
+    for c0 in (0, 3, 5):
+        x = np.exp(2j * np.pi * (c0 / N) * m).astype(np.complex64)
+        ch = o.pfb_channelizer(N, N // 2, taps, x, 4096 * 2 // N)
+        e = (np.abs(ch[:, 20:]) ** 2).sum(axis=1)
+        assert e[c0] / e.sum() > 0.9, (c0, e / e.sum())
+
+
 def test_pfb_channelizer_linearity(oracle_lib):
     o = oracle_lib
     N = 4
