@@ -258,7 +258,8 @@ def measure_config3(fa, lib, torch):
                 "unit": "TFLOP/s",
                 "frac": round(rs_tf / FP32_PEAK_TFLOPS, 4),
                 "traffic": None,
-                "kernel": "k_resamp_tiled_cf32",
+                "kernel": "k_decim4_mfma_tpl via the 1:4 resampler route "
+                          "(k_resamp_tiled_cf32 for interp>1)",
                 "ms_per_launch": round(rs_ms, 4),
             },
         }

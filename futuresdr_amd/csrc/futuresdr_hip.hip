@@ -694,6 +694,11 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
     const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
     const unsigned SPm = (elemsP + 31u) & ~31u;
+    /* each phase plane is split into 4 sub-planes by element%4 so a
+     * lane's MFMA K-walk (idx = ab+4s) becomes stride-1: one
+     * ds_read_b128 feeds 4 K-steps (the b32-per-MFMA A-reads were the
+     * round-1 issue-stall, SQ_WAIT_INST_ANY 46% — profiles/pmc_r02) */
+    const unsigned SUB = SPm / 4;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     /* two phases resident at a time ([re_v0, re_v1, im_v0, im_v1]) —
      * halves LDS vs all-phase planes, doubling resident blocks/CU;
@@ -736,24 +741,27 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
             unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
             unsigned i = idx >> 1, vloc = idx & 1u;
             if (i < elemsP) {
-                unsigned d = mfma_swz(i);
+                unsigned d = (i & 3u) * SUB + (i >> 2);
                 planes[vloc * SPm + d] = stg[j].x;
                 planes[(2 + vloc) * SPm + d] = stg[j].y;
             }
         }
     };
-    const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-    /* Two interleaved accumulator pairs per half: a single C chain can
-     * only issue a dependent MFMA every result latency (~2x the issue
-     * rate), stalling the SIMD half the time — the round-1 WAIT_ANY
-     * ~50%. Even/odd K-steps accumulate independently per phase and
-     * merge at the output (fp32 reassociation, within tolerance). */
+    /* lane (r16,k4)'s K-walk in a sub-plane: sub = k4, dword offset
+     * wave*64 + 4*r16 + s — stride-1 in s, 16 B aligned at s%4==0 */
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+    /* Two interleaved accumulator pairs + b128 A-reads: each
+     * ds_read_b128 pair feeds 8 MFMAs (4 K-steps x re/im), cutting the
+     * per-MFMA issue overhead that stalled round 1 (SQ_WAIT_INST_ANY).
+     * Even/odd K-steps accumulate independently per phase and merge at
+     * the output (fp32 reassociation, within tolerance). */
     auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
                          v4f& cim1) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
-            const float* pre = planes + (unsigned)vloc * SPm;
-            const float* pim = planes + (unsigned)(2 + vloc) * SPm;
+            const float* pre = planes + (unsigned)vloc * SPm + asub;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm + asub;
             const int v = 2 * h + vloc;
             float bfrag[KKD / 4];
 #pragma unroll
@@ -763,24 +771,30 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
                                               FFT/staging phases of
                                               co-resident blocks (T5) */
 #pragma unroll
-            for (int s = 0; s < KKD / 4 - 1; s += 2) {
-                float a_re0 = pre[mfma_swz(ab + 4 * s)];
-                float a_im0 = pim[mfma_swz(ab + 4 * s)];
-                float a_re1 = pre[mfma_swz(ab + 4 * (s + 1))];
-                float a_im1 = pim[mfma_swz(ab + 4 * (s + 1))];
+            for (int t = 0; t < (KKD / 4) / 4; t++) {
+                float4 ar = *(const float4*)&pre[abase + 4 * t];
+                float4 ai = *(const float4*)&pim[abase + 4 * t];
                 cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re0, bfrag[s], cre0, 0, 0, 0);
+                    ar.x, bfrag[4 * t], cre0, 0, 0, 0);
                 cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im0, bfrag[s], cim0, 0, 0, 0);
+                    ai.x, bfrag[4 * t], cim0, 0, 0, 0);
                 cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re1, bfrag[s + 1], cre1, 0, 0, 0);
+                    ar.y, bfrag[4 * t + 1], cre1, 0, 0, 0);
                 cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im1, bfrag[s + 1], cim1, 0, 0, 0);
+                    ai.y, bfrag[4 * t + 1], cim1, 0, 0, 0);
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim0, 0, 0, 0);
+                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre1, 0, 0, 0);
+                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim1, 0, 0, 0);
             }
-            if (KKD / 4 & 1) {
-                const int s = KKD / 4 - 1;
-                float a_re = pre[mfma_swz(ab + 4 * s)];
-                float a_im = pim[mfma_swz(ab + 4 * s)];
+#pragma unroll
+            for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
+                float a_re = pre[abase + s];
+                float a_im = pim[abase + s];
                 cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
                     a_re, bfrag[s], cre0, 0, 0, 0);
                 cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
@@ -829,6 +843,11 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
     static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
     const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
     const unsigned SPm = (elemsP + 31u) & ~31u;
+    /* each phase plane is split into 4 sub-planes by element%4 so a
+     * lane's MFMA K-walk (idx = ab+4s) becomes stride-1: one
+     * ds_read_b128 feeds 4 K-steps (the b32-per-MFMA A-reads were the
+     * round-1 issue-stall, SQ_WAIT_INST_ANY 46% — profiles/pmc_r02) */
+    const unsigned SUB = SPm / 4;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     /* two phases resident at a time ([re_v0, re_v1, im_v0, im_v1]) —
      * halves LDS vs all-phase planes, doubling resident blocks/CU;
@@ -871,24 +890,27 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
             unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
             unsigned i = idx >> 1, vloc = idx & 1u;
             if (i < elemsP) {
-                unsigned d = mfma_swz(i);
+                unsigned d = (i & 3u) * SUB + (i >> 2);
                 planes[vloc * SPm + d] = stg[j].x;
                 planes[(2 + vloc) * SPm + d] = stg[j].y;
             }
         }
     };
-    const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-    /* Two interleaved accumulator pairs per half: a single C chain can
-     * only issue a dependent MFMA every result latency (~2x the issue
-     * rate), stalling the SIMD half the time — the round-1 WAIT_ANY
-     * ~50%. Even/odd K-steps accumulate independently per phase and
-     * merge at the output (fp32 reassociation, within tolerance). */
+    /* lane (r16,k4)'s K-walk in a sub-plane: sub = k4, dword offset
+     * wave*64 + 4*r16 + s — stride-1 in s, 16 B aligned at s%4==0 */
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+    /* Two interleaved accumulator pairs + b128 A-reads: each
+     * ds_read_b128 pair feeds 8 MFMAs (4 K-steps x re/im), cutting the
+     * per-MFMA issue overhead that stalled round 1 (SQ_WAIT_INST_ANY).
+     * Even/odd K-steps accumulate independently per phase and merge at
+     * the output (fp32 reassociation, within tolerance). */
     auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
                          v4f& cim1) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
-            const float* pre = planes + (unsigned)vloc * SPm;
-            const float* pim = planes + (unsigned)(2 + vloc) * SPm;
+            const float* pre = planes + (unsigned)vloc * SPm + asub;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm + asub;
             const int v = 2 * h + vloc;
             float bfrag[KKD / 4];
 #pragma unroll
@@ -898,24 +920,30 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
                                               FFT/staging phases of
                                               co-resident blocks (T5) */
 #pragma unroll
-            for (int s = 0; s < KKD / 4 - 1; s += 2) {
-                float a_re0 = pre[mfma_swz(ab + 4 * s)];
-                float a_im0 = pim[mfma_swz(ab + 4 * s)];
-                float a_re1 = pre[mfma_swz(ab + 4 * (s + 1))];
-                float a_im1 = pim[mfma_swz(ab + 4 * (s + 1))];
+            for (int t = 0; t < (KKD / 4) / 4; t++) {
+                float4 ar = *(const float4*)&pre[abase + 4 * t];
+                float4 ai = *(const float4*)&pim[abase + 4 * t];
                 cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re0, bfrag[s], cre0, 0, 0, 0);
+                    ar.x, bfrag[4 * t], cre0, 0, 0, 0);
                 cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im0, bfrag[s], cim0, 0, 0, 0);
+                    ai.x, bfrag[4 * t], cim0, 0, 0, 0);
                 cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re1, bfrag[s + 1], cre1, 0, 0, 0);
+                    ar.y, bfrag[4 * t + 1], cre1, 0, 0, 0);
                 cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im1, bfrag[s + 1], cim1, 0, 0, 0);
+                    ai.y, bfrag[4 * t + 1], cim1, 0, 0, 0);
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim0, 0, 0, 0);
+                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre1, 0, 0, 0);
+                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim1, 0, 0, 0);
             }
-            if (KKD / 4 & 1) {
-                const int s = KKD / 4 - 1;
-                float a_re = pre[mfma_swz(ab + 4 * s)];
-                float a_im = pim[mfma_swz(ab + 4 * s)];
+#pragma unroll
+            for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
+                float a_re = pre[abase + s];
+                float a_im = pim[abase + s];
                 cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
                     a_re, bfrag[s], cre0, 0, 0, 0);
                 cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
@@ -986,6 +1014,11 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
     static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
     const unsigned elemsP = 2048 + KKD + 8;     /* per phase plane */
     const unsigned SPm = (elemsP + 31u) & ~31u;
+    /* each phase plane is split into 4 sub-planes by element%4 so a
+     * lane's MFMA K-walk (idx = ab+4s) becomes stride-1: one
+     * ds_read_b128 feeds 4 K-steps (the b32-per-MFMA A-reads were the
+     * round-1 issue-stall, SQ_WAIT_INST_ANY 46% — profiles/pmc_r02) */
+    const unsigned SUB = SPm / 4;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     /* two phases resident at a time ([re_v0, re_v1, im_v0, im_v1]) —
      * halves LDS vs all-phase planes, doubling resident blocks/CU;
@@ -1028,24 +1061,27 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
             unsigned idx = (unsigned)(tid + j * 512);
             unsigned i = idx >> 1, vloc = idx & 1u;
             if (i < elemsP) {
-                unsigned d = mfma_swz(i);
+                unsigned d = (i & 3u) * SUB + (i >> 2);
                 planes[vloc * SPm + d] = stg[j].x;
                 planes[(2 + vloc) * SPm + d] = stg[j].y;
             }
         }
     };
-    const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-    /* Two interleaved accumulator pairs per half: a single C chain can
-     * only issue a dependent MFMA every result latency (~2x the issue
-     * rate), stalling the SIMD half the time — the round-1 WAIT_ANY
-     * ~50%. Even/odd K-steps accumulate independently per phase and
-     * merge at the output (fp32 reassociation, within tolerance). */
+    /* lane (r16,k4)'s K-walk in a sub-plane: sub = k4, dword offset
+     * wave*64 + 4*r16 + s — stride-1 in s, 16 B aligned at s%4==0 */
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+    /* Two interleaved accumulator pairs + b128 A-reads: each
+     * ds_read_b128 pair feeds 8 MFMAs (4 K-steps x re/im), cutting the
+     * per-MFMA issue overhead that stalled round 1 (SQ_WAIT_INST_ANY).
+     * Even/odd K-steps accumulate independently per phase and merge at
+     * the output (fp32 reassociation, within tolerance). */
     auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
                          v4f& cim1) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
-            const float* pre = planes + (unsigned)vloc * SPm;
-            const float* pim = planes + (unsigned)(2 + vloc) * SPm;
+            const float* pre = planes + (unsigned)vloc * SPm + asub;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm + asub;
             const int v = 2 * h + vloc;
             float bfrag[KKD / 4];
 #pragma unroll
@@ -1055,24 +1091,30 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
                                               FFT/staging phases of
                                               co-resident blocks (T5) */
 #pragma unroll
-            for (int s = 0; s < KKD / 4 - 1; s += 2) {
-                float a_re0 = pre[mfma_swz(ab + 4 * s)];
-                float a_im0 = pim[mfma_swz(ab + 4 * s)];
-                float a_re1 = pre[mfma_swz(ab + 4 * (s + 1))];
-                float a_im1 = pim[mfma_swz(ab + 4 * (s + 1))];
+            for (int t = 0; t < (KKD / 4) / 4; t++) {
+                float4 ar = *(const float4*)&pre[abase + 4 * t];
+                float4 ai = *(const float4*)&pim[abase + 4 * t];
                 cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re0, bfrag[s], cre0, 0, 0, 0);
+                    ar.x, bfrag[4 * t], cre0, 0, 0, 0);
                 cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im0, bfrag[s], cim0, 0, 0, 0);
+                    ai.x, bfrag[4 * t], cim0, 0, 0, 0);
                 cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re1, bfrag[s + 1], cre1, 0, 0, 0);
+                    ar.y, bfrag[4 * t + 1], cre1, 0, 0, 0);
                 cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im1, bfrag[s + 1], cim1, 0, 0, 0);
+                    ai.y, bfrag[4 * t + 1], cim1, 0, 0, 0);
+                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre0, 0, 0, 0);
+                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim0, 0, 0, 0);
+                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre1, 0, 0, 0);
+                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim1, 0, 0, 0);
             }
-            if (KKD / 4 & 1) {
-                const int s = KKD / 4 - 1;
-                float a_re = pre[mfma_swz(ab + 4 * s)];
-                float a_im = pim[mfma_swz(ab + 4 * s)];
+#pragma unroll
+            for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
+                float a_re = pre[abase + s];
+                float a_im = pim[abase + s];
                 cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
                     a_re, bfrag[s], cre0, 0, 0, 0);
                 cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
