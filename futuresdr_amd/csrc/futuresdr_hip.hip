@@ -1957,13 +1957,33 @@ __device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
 
 __global__ void k_fill_uniform_cf32(float2* __restrict__ out, long long n,
                                     uint64_t seed, uint64_t offset) {
+    /* pair of samples per iteration, one float4 store (the source write
+     * is inside the bench's timed region — keep it at write bandwidth) */
+    long long np = n >> 1;
     long long stride = (long long)gridDim.x * blockDim.x;
-    for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-         i < n; i += stride) {
-        uint64_t h = splitmix64(seed ^ (0x5D5D5D5Dull + offset + (uint64_t)i));
-        uint32_t lo = (uint32_t)h, hi = (uint32_t)(h >> 32);
-        out[i] = make_float2((lo >> 8) * (2.0f / 16777216.0f) - 1.0f,
-                             (hi >> 8) * (2.0f / 16777216.0f) - 1.0f);
+    for (long long p = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         p < np; p += stride) {
+        long long i = 2 * p;
+        uint64_t h0 =
+            splitmix64(seed ^ (0x5D5D5D5Dull + offset + (uint64_t)i));
+        uint64_t h1 =
+            splitmix64(seed ^ (0x5D5D5D5Dull + offset + (uint64_t)i + 1));
+        const float s = 2.0f / 16777216.0f;
+        float4 v = make_float4(((uint32_t)h0 >> 8) * s - 1.0f,
+                               ((uint32_t)(h0 >> 32) >> 8) * s - 1.0f,
+                               ((uint32_t)h1 >> 8) * s - 1.0f,
+                               ((uint32_t)(h1 >> 32) >> 8) * s - 1.0f);
+        *(float4*)&out[i] = v;
+    }
+    if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+        long long i = n - 1;
+        uint64_t h =
+            splitmix64(seed ^ (0x5D5D5D5Dull + offset + (uint64_t)i));
+        out[i] = make_float2(((uint32_t)h >> 8) * (2.0f / 16777216.0f) -
+                                 1.0f,
+                             ((uint32_t)(h >> 32) >> 8) *
+                                     (2.0f / 16777216.0f) -
+                                 1.0f);
     }
 }
 
@@ -3139,9 +3159,9 @@ extern "C" int fsdr_fill_uniform_cf32(void* d_ptr, size_t n, uint64_t seed,
                                       uint64_t offset, void* stream) {
     REQUIRE_GPU();
     hipLaunchKernelGGL(k_fill_uniform_cf32,
-                       dim3(grid_for((long long)n, 256)), dim3(256), 0,
-                       (hipStream_t)stream, (float2*)d_ptr, (long long)n,
-                       seed, offset);
+                       dim3(grid_for((long long)(n / 2 + 1), 256)),
+                       dim3(256), 0, (hipStream_t)stream, (float2*)d_ptr,
+                       (long long)n, seed, offset);
     HIP_TRY(hipGetLastError());
     return FSDR_OK;
 }

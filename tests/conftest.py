@@ -7,6 +7,15 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
 
+# Import torch (its bundled HIP runtime) BEFORE any test dlopens the
+# futuresdr_amd C-ABI library: loading /opt/rocm's libamdhip64 first makes
+# torch's lazy HIP init report no devices (see bench.py header). Harmless
+# on CPU-only boxes.
+try:
+    import torch  # noqa: F401
+except ImportError:
+    pass
+
 
 def pytest_configure(config):
     config.addinivalue_line(
