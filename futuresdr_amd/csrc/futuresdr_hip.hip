@@ -831,8 +831,8 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     }
 }
 
-template <int KKD>
-__global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_tpl(
+template <int KKD, int MINWG = 1>
+__global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
     const float2* __restrict__ in, float2* __restrict__ out,
     const float* __restrict__ rtv /* [4][KKD], rtv[v][u] = rt[4u+v] */,
     long long n_out, long long n_in_valid,
@@ -3641,6 +3641,19 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                                (long long)n_in,
                                (const float2*)c->fft->d_twid,
                                (float*)d_mag);
+            HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
+        const char* occ8 = getenv("FSDR_CHAIN_OCC8");
+        if (occ8 && atoi(occ8) != 0 && KK == 80) {
+            /* experiment: force 8 waves/SIMD occupancy (VGPR capped to
+             * 64 by the compiler; some spill risk) */
+            hipLaunchKernelGGL(
+                HIP_KERNEL_NAME((k_decim4_fft_mfma_tpl<80, 8>)),
+                dim3(grid), dim3(MDFIR_BLOCK), lds, st,
+                (const float2*)d_in, spec_dst, c->fused->d_mtaps,
+                (long long)prod, (long long)n_in,
+                (const float2*)c->fft->d_twid, (float*)d_mag, (int)L);
             HIP_TRY(hipGetLastError());
             return FSDR_OK;
         }
