@@ -528,44 +528,26 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
             load_tile(tile + gridDim.x);
 
         const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
-        /* two interleaved accumulator pairs: a single C chain issues one
-         * dependent MFMA per result latency (~2x the 8-clk issue rate),
-         * stalling the SIMD half the time — the round-1 WAIT_ANY ~50%.
-         * Even/odd K-steps accumulate independently and merge at the end
-         * (fp32 reassociation, within the parity tolerance). */
-        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
-        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
+        /* one accumulator pair, re/im alternating: same-C spacing = 2
+         * MFMA issues = 64 cyc/SIMD >= the 40-cyc dependent latency
+         * (extra pairs measured slower — register pressure only) */
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int s = 0; s < KK / 4 - 1; s += 2) {
-            float a_re0 = s_re[mfma_swz(ab + 4 * s)];
-            float a_im0 = s_im[mfma_swz(ab + 4 * s)];
-            float a_re1 = s_re[mfma_swz(ab + 4 * (s + 1))];
-            float a_im1 = s_im[mfma_swz(ab + 4 * (s + 1))];
-            cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re0, bfrag[s],
-                                                        cre0, 0, 0, 0);
-            cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im0, bfrag[s],
-                                                        cim0, 0, 0, 0);
-            cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re1, bfrag[s + 1],
-                                                        cre1, 0, 0, 0);
-            cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im1, bfrag[s + 1],
-                                                        cim1, 0, 0, 0);
-        }
-        if (KK / 4 & 1) {
-            const int s = KK / 4 - 1;
+        for (int s = 0; s < KK / 4; s++) {
             float a_re = s_re[mfma_swz(ab + 4 * s)];
             float a_im = s_im[mfma_swz(ab + 4 * s)];
-            cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s],
-                                                        cre0, 0, 0, 0);
-            cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s],
-                                                        cim0, 0, 0, 0);
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
+                                                       0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
+                                                       0, 0, 0);
         }
         /* C layout: col = lane&15, row = (lane>>4)*4 + q (cdna4 16x16) */
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
             long long o = out_base + (long long)wave * 256 + 16 * row + r16;
-            if (o < n_out)
-                out[o] = make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
+            if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
         }
         __syncthreads();
     }
@@ -751,13 +733,14 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
      * wave*64 + 4*r16 + s — stride-1 in s, 16 B aligned at s%4==0 */
     const unsigned abase = (unsigned)wave * 64 + 4u * r16;
     const unsigned asub = (unsigned)k4 * SUB;
-    /* Two interleaved accumulator pairs + b128 A-reads: each
-     * ds_read_b128 pair feeds 8 MFMAs (4 K-steps x re/im), cutting the
-     * per-MFMA issue overhead that stalled round 1 (SQ_WAIT_INST_ANY).
-     * Even/odd K-steps accumulate independently per phase and merge at
-     * the output (fp32 reassociation, within tolerance). */
-    auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
-                         v4f& cim1) {
+    /* b128 A-reads: one ds_read_b128 pair feeds 8 MFMAs (4 K-steps x
+     * re/im), cutting the per-MFMA issue overhead (round-1 b32 reads:
+     * SQ_WAIT_INST_ANY-heavy, profiles/pmc_sq_r02.txt). One accumulator
+     * pair, re/im alternating: same-C spacing = 2 MFMA issues = 64
+     * cyc/SIMD >= the 40-cyc dependent latency, so extra accumulator
+     * pairs only cost registers (measured: dual pairs were ~3% slower,
+     * forcing 8 waves/SIMD via VGPR caps 2x slower from spills). */
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
             const float* pre = planes + (unsigned)vloc * SPm + asub;
@@ -774,31 +757,31 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
             for (int t = 0; t < (KKD / 4) / 4; t++) {
                 float4 ar = *(const float4*)&pre[abase + 4 * t];
                 float4 ai = *(const float4*)&pim[abase + 4 * t];
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.x, bfrag[4 * t], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.x, bfrag[4 * t], cim0, 0, 0, 0);
-                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.y, bfrag[4 * t + 1], cre1, 0, 0, 0);
-                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.y, bfrag[4 * t + 1], cim1, 0, 0, 0);
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.z, bfrag[4 * t + 2], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.z, bfrag[4 * t + 2], cim0, 0, 0, 0);
-                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.w, bfrag[4 * t + 3], cre1, 0, 0, 0);
-                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.w, bfrag[4 * t + 3], cim1, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.x, bfrag[4 * t], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.x, bfrag[4 * t], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
             }
 #pragma unroll
             for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
                 float a_re = pre[abase + s];
                 float a_im = pim[abase + s];
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[s], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[s], cim0, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
         }
@@ -808,24 +791,24 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_mfma_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * MDFIR_TILE;
-        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
-        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
         load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
-        mfma_half(0, cre0, cim0, cre1, cim1);
+        mfma_half(0, cre, cim);
         __syncthreads();
         write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
-        mfma_half(1, cre0, cim0, cre1, cim1);
+        mfma_half(1, cre, cim);
 #pragma unroll
         for (int q = 0; q < 4; q++) {
             int row = k4 * 4 + q;
             long long o = out_base + (long long)wave * 256 + 16 * row + r16;
             if (o < n_out)
-                out[o] = make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
+                out[o] = make_float2(cre[q], cim[q]);
         }
         __syncthreads();
     }
@@ -900,13 +883,14 @@ __global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
      * wave*64 + 4*r16 + s — stride-1 in s, 16 B aligned at s%4==0 */
     const unsigned abase = (unsigned)wave * 64 + 4u * r16;
     const unsigned asub = (unsigned)k4 * SUB;
-    /* Two interleaved accumulator pairs + b128 A-reads: each
-     * ds_read_b128 pair feeds 8 MFMAs (4 K-steps x re/im), cutting the
-     * per-MFMA issue overhead that stalled round 1 (SQ_WAIT_INST_ANY).
-     * Even/odd K-steps accumulate independently per phase and merge at
-     * the output (fp32 reassociation, within tolerance). */
-    auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
-                         v4f& cim1) {
+    /* b128 A-reads: one ds_read_b128 pair feeds 8 MFMAs (4 K-steps x
+     * re/im), cutting the per-MFMA issue overhead (round-1 b32 reads:
+     * SQ_WAIT_INST_ANY-heavy, profiles/pmc_sq_r02.txt). One accumulator
+     * pair, re/im alternating: same-C spacing = 2 MFMA issues = 64
+     * cyc/SIMD >= the 40-cyc dependent latency, so extra accumulator
+     * pairs only cost registers (measured: dual pairs were ~3% slower,
+     * forcing 8 waves/SIMD via VGPR caps 2x slower from spills). */
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
             const float* pre = planes + (unsigned)vloc * SPm + asub;
@@ -923,31 +907,31 @@ __global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
             for (int t = 0; t < (KKD / 4) / 4; t++) {
                 float4 ar = *(const float4*)&pre[abase + 4 * t];
                 float4 ai = *(const float4*)&pim[abase + 4 * t];
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.x, bfrag[4 * t], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.x, bfrag[4 * t], cim0, 0, 0, 0);
-                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.y, bfrag[4 * t + 1], cre1, 0, 0, 0);
-                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.y, bfrag[4 * t + 1], cim1, 0, 0, 0);
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.z, bfrag[4 * t + 2], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.z, bfrag[4 * t + 2], cim0, 0, 0, 0);
-                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.w, bfrag[4 * t + 3], cre1, 0, 0, 0);
-                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.w, bfrag[4 * t + 3], cim1, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.x, bfrag[4 * t], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.x, bfrag[4 * t], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
             }
 #pragma unroll
             for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
                 float a_re = pre[abase + s];
                 float a_im = pim[abase + s];
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[s], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[s], cim0, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
         }
@@ -957,18 +941,18 @@ __global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
         const long long out_base = tile * MDFIR_TILE;
-        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
-        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
         load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
-        mfma_half(0, cre0, cim0, cre1, cim1);
+        mfma_half(0, cre, cim);
         __syncthreads();
         write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
-        mfma_half(1, cre0, cim0, cre1, cim1);
+        mfma_half(1, cre, cim);
         __syncthreads(); /* phase planes are dead; reuse them as FFT LDS */
         float2* ping = (float2*)planes;       /* 1024 float2 = 8 KB */
         float2* pong = ping + 1024;           /* fits in 4*SPm floats */
@@ -978,7 +962,7 @@ __global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
             int row = k4 * 4 + q;
             unsigned pos = wave * 256 + 16 * row + r16; /* y2 idx in tile */
             ping[(pos & ~Lm) | fft_swz(pos & Lm)] =
-                make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
+                make_float2(cre[q], cim[q]);
         }
         __syncthreads();
         { /* one FFT per frame; frames share the block in lockstep */
@@ -1071,13 +1055,14 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
      * wave*64 + 4*r16 + s — stride-1 in s, 16 B aligned at s%4==0 */
     const unsigned abase = (unsigned)wave * 64 + 4u * r16;
     const unsigned asub = (unsigned)k4 * SUB;
-    /* Two interleaved accumulator pairs + b128 A-reads: each
-     * ds_read_b128 pair feeds 8 MFMAs (4 K-steps x re/im), cutting the
-     * per-MFMA issue overhead that stalled round 1 (SQ_WAIT_INST_ANY).
-     * Even/odd K-steps accumulate independently per phase and merge at
-     * the output (fp32 reassociation, within tolerance). */
-    auto mfma_half = [&](int h, v4f& cre0, v4f& cim0, v4f& cre1,
-                         v4f& cim1) {
+    /* b128 A-reads: one ds_read_b128 pair feeds 8 MFMAs (4 K-steps x
+     * re/im), cutting the per-MFMA issue overhead (round-1 b32 reads:
+     * SQ_WAIT_INST_ANY-heavy, profiles/pmc_sq_r02.txt). One accumulator
+     * pair, re/im alternating: same-C spacing = 2 MFMA issues = 64
+     * cyc/SIMD >= the 40-cyc dependent latency, so extra accumulator
+     * pairs only cost registers (measured: dual pairs were ~3% slower,
+     * forcing 8 waves/SIMD via VGPR caps 2x slower from spills). */
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
 #pragma unroll
         for (int vloc = 0; vloc < 2; vloc++) {
             const float* pre = planes + (unsigned)vloc * SPm + asub;
@@ -1094,31 +1079,31 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
             for (int t = 0; t < (KKD / 4) / 4; t++) {
                 float4 ar = *(const float4*)&pre[abase + 4 * t];
                 float4 ai = *(const float4*)&pim[abase + 4 * t];
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.x, bfrag[4 * t], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.x, bfrag[4 * t], cim0, 0, 0, 0);
-                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.y, bfrag[4 * t + 1], cre1, 0, 0, 0);
-                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.y, bfrag[4 * t + 1], cim1, 0, 0, 0);
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.z, bfrag[4 * t + 2], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.z, bfrag[4 * t + 2], cim0, 0, 0, 0);
-                cre1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ar.w, bfrag[4 * t + 3], cre1, 0, 0, 0);
-                cim1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    ai.w, bfrag[4 * t + 3], cim1, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.x, bfrag[4 * t], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.x, bfrag[4 * t], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
             }
 #pragma unroll
             for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
                 float a_re = pre[abase + s];
                 float a_im = pim[abase + s];
-                cre0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_re, bfrag[s], cre0, 0, 0, 0);
-                cim0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                    a_im, bfrag[s], cim0, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
             }
             __builtin_amdgcn_s_setprio(0);
         }
@@ -1128,18 +1113,18 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
     for (long long tile = blockIdx.x;
          tile * (long long)2048 < n_out; tile += gridDim.x) {
         const long long out_base = tile * 2048;
-        v4f cre0 = {0.f, 0.f, 0.f, 0.f}, cim0 = {0.f, 0.f, 0.f, 0.f};
-        v4f cre1 = {0.f, 0.f, 0.f, 0.f}, cim1 = {0.f, 0.f, 0.f, 0.f};
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
         write_half(stgA);
         __syncthreads();
         load_half(tile, 1, stgB);     /* in flight under half-0 MFMAs */
-        mfma_half(0, cre0, cim0, cre1, cim1);
+        mfma_half(0, cre, cim);
         __syncthreads();
         write_half(stgB);
         __syncthreads();
         if ((tile + gridDim.x) * (long long)2048 < n_out)
             load_half(tile + gridDim.x, 0, stgA); /* under half-1 MFMAs */
-        mfma_half(1, cre0, cim0, cre1, cim1);
+        mfma_half(1, cre, cim);
         __syncthreads(); /* phase planes are dead; reuse them as FFT LDS:
                             frame f ping = fbase + f*2048, pong = +1024 */
         float2* fbase = (float2*)planes; /* 4096 float2 = 32 KB */
@@ -1149,7 +1134,7 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
             int pos = wave * 256 + 16 * row + r16; /* y2 index in tile */
             int fr = pos >> 10, idx = pos & 1023;
             fbase[fr * 2048 + fft_swz((unsigned)idx)] =
-                make_float2(cre0[q] + cre1[q], cim0[q] + cim1[q]);
+                make_float2(cre[q], cim[q]);
         }
         __syncthreads();
         {
