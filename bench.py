@@ -52,8 +52,10 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--samples", type=int, default=1 << 28,
-                   help="chain-input Complex32 samples per step per GPU")
+    p.add_argument("--samples", type=int, default=1 << 30,
+                   help="chain-input Complex32 samples per step per GPU "
+                        "(288 GB HBM3E: 8.6 GB batches amortize launch "
+                        "overheads — profiles/throughput_curve_r02.txt)")
     p.add_argument("--fft", type=int, default=1024)
     p.add_argument("--decim", type=int, default=4)
     p.add_argument("--taps", type=int, default=127)
@@ -65,6 +67,10 @@ def parse_args():
                    help="skip the PCIe-fed ring streaming leg")
     p.add_argument("--skip-config3", action="store_true",
                    help="skip the config-3 (FM resampler chain) leg")
+    p.add_argument("--skip-roofline", action="store_true",
+                   help="skip the roofline reps (for PMC traffic passes "
+                        "whose counters the extra d_null-writing reps "
+                        "would pollute)")
     p.add_argument("--streaming-chunk", type=int, default=1 << 22,
                    help="ring chunk size in samples for the streaming leg")
     p.add_argument("--streaming-chunks", type=int, default=64,
@@ -405,9 +411,11 @@ def main():
     value = n_gpus * S * args.steps / elapsed / 1e6  # whole-job MSample/s
 
     if rank == 0:
-        roofline = measure_chain_roofline(fa, torch, chain, d_in, S,
-                                          taps1, taps2, args.decim,
-                                          args.fft, args.traffic_file)
+        roofline = None
+        if not args.skip_roofline:
+            roofline = measure_chain_roofline(fa, torch, chain, d_in, S,
+                                              taps1, taps2, args.decim,
+                                              args.fft, args.traffic_file)
         cpu_baseline = None
         if n_gpus == 1 and not args.skip_cpu_baseline:
             log("measuring CPU baseline (oracle chain, all cores)...")

@@ -3581,16 +3581,7 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
     if (consumed) *consumed = prod * D; /* chain-input samples per frame set */
     if (produced) *produced = prod;
     if (frames == 0) return FSDR_OK;
-    int rc = ensure_dev((void**)&c->d_y2, &c->y2_cap,
-                        (prod + 8) * sizeof(float2));
-    if (rc) return rc;
-    float2* out2 = (float2*)d_out;
-    if (!out2) {
-        rc = ensure_dev((void**)&c->d_null, &c->null_cap,
-                        (prod + 8) * sizeof(float2));
-        if (rc) return rc;
-        out2 = c->d_null;
-    }
+    int rc;
     const char* ffz = getenv("FSDR_CHAIN_FFTFUSE");
     const bool fft_fusable =
         (L == 64 || L == 128 || L == 256 || L == 512 || L == 1024);
@@ -3610,7 +3601,12 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                       4 * ((size_t)KK + 16)) * sizeof(float);
         float2* spec_dst = (float2*)d_out; /* null + mag-only: skip the
                                               discarded spectra write */
-        if (!spec_dst && !d_mag) spec_dst = out2;
+        if (!spec_dst && !d_mag) { /* NullSink scratch keeps work observable */
+            rc = ensure_dev((void**)&c->d_null, &c->null_cap,
+                            (prod + 8) * sizeof(float2));
+            if (rc) return rc;
+            spec_dst = c->d_null;
+        }
         const char* b512 = getenv("FSDR_CHAIN_BLOCK512");
         if (b512 && atoi(b512) != 0 && L == 1024 && KK == 80 &&
             prod % 2048 == 0) {
@@ -3665,6 +3661,16 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
 #undef CHAIN_TPL_CASE
         HIP_TRY(hipGetLastError());
         return FSDR_OK;
+    }
+    rc = ensure_dev((void**)&c->d_y2, &c->y2_cap,
+                    (prod + 8) * sizeof(float2));
+    if (rc) return rc;
+    float2* out2 = (float2*)d_out;
+    if (!out2) {
+        rc = ensure_dev((void**)&c->d_null, &c->null_cap,
+                        (prod + 8) * sizeof(float2));
+        if (rc) return rc;
+        out2 = c->d_null;
     }
     if (c->fused) {
         rc = launch_decim_cf32(c->fused, d_in, c->d_y2, prod, n_in, st);
