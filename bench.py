@@ -78,7 +78,7 @@ def parse_args():
     p.add_argument("--cpu-sample", type=int, default=0,
                    help="fixed CPU-baseline sample size (0 = auto ~10s)")
     p.add_argument("--traffic-file", default=os.path.join(
-        REPO, "profiles", "traffic_r01.json"))
+        REPO, "profiles", "traffic_r02.json"))
     return p.parse_args()
 
 
