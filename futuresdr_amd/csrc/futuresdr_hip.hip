@@ -1813,6 +1813,60 @@ __device__ void fft1024_block(float2* ping, float2* pong,
     /* 5 radix-4 stages = 5 ping/pong swaps: result in `pong`. */
 }
 
+/* ---- Bluestein (non-pow2 lengths) ------------------------------------ *
+ * X[k] = w(k) * IFFT_M(FFT_M(x.w) * B)[k] / M, w the quadratic chirp
+ * and B the precomputed transform of the wrapped conj chirp (tables in
+ * fsdr_fft_cf32_create). The M-point passes reuse k_fft_stockham. */
+__global__ void k_bluestein_pre(const float2* __restrict__ in,
+                                float2* __restrict__ out /* frames*M */,
+                                const float2* __restrict__ chirp, int n,
+                                int M, long long frames, int shift_in) {
+    long long total = frames * M;
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         id < total; id += stride) {
+        long long fr = id / M;
+        int j = (int)(id - fr * M);
+        float2 v = make_float2(0.f, 0.f);
+        if (j < n) {
+            int src = shift_in ? (j + n / 2) % n : j; /* fft.rs:179-185 */
+            v = cmulf(in[fr * n + src], chirp[j]);
+        }
+        out[id] = v;
+    }
+}
+
+__global__ void k_bluestein_mul(float2* __restrict__ x,
+                                const float2* __restrict__ B, int M,
+                                long long frames) {
+    long long total = frames * M;
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         id < total; id += stride)
+        x[id] = cmulf(x[id], B[id % M]);
+}
+
+__global__ void k_bluestein_post(const float2* __restrict__ conv,
+                                 float2* __restrict__ out,
+                                 float* __restrict__ mag /* nullable */,
+                                 const float2* __restrict__ chirp, int n,
+                                 int M, long long frames, int shift_out,
+                                 float scale) {
+    long long total = frames * n;
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long id = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+         id < total; id += stride) {
+        long long fr = id / n;
+        int k = (int)(id - fr * n);
+        int src = shift_out ? (k + n / 2) % n : k; /* fft.rs:196-204 */
+        float2 v = cmulf(conv[fr * M + src], chirp[src]);
+        v.x *= scale;
+        v.y *= scale;
+        out[id] = v;
+        if (mag) mag[id] = v.x * v.x + v.y * v.y;
+    }
+}
+
 __global__ __launch_bounds__(256) void k_fft_stockham(
     const float2* __restrict__ in, float2* __restrict__ out,
     float* __restrict__ mag_out /* nullable */,
@@ -2054,6 +2108,13 @@ struct fsdr_filter {
     size_t fft_len = 0;
     int inverse = 0, fft_shift = 0;
     float norm = 0.f;
+    /* Bluestein (non-pow2 fft_len): M = next pow2 >= 2*len-1; chirp
+     * w(k) = e^{sigma*pi*i*k^2/len}; B = DFT_M of the wrapped conj
+     * chirp. fft.rs is generic over rustfft plan lengths; this closes
+     * the pow2-only gap via two pow2 M-point passes per batch. */
+    size_t fft_m = 0;
+    float2* d_chirp = nullptr;
+    float2* d_B = nullptr;
     int n_taps_padded = 0;   /* device taps length (leading zeros) */
     float* d_taps = nullptr;
     int tp_tpl = 0;          /* template tap count (reversed taps) or 0 */
@@ -2311,8 +2372,10 @@ extern "C" fsdr_filter* fsdr_resamp_cf32_create(size_t interp, size_t decim,
 extern "C" fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse,
                                              int fft_shift,
                                              const float* normalize) {
-    if (len < 4 || len > 4096 || (len & (len - 1)) != 0) {
-        set_err("fft len must be a power of two in [4,4096]");
+    const bool pow2 = (len & (len - 1)) == 0;
+    if (pow2 ? (len < 4 || len > 4096) : (len < 2 || len > 2048)) {
+        set_err("fft len: pow2 in [4,4096], or any length in [2,2048] "
+                "(Bluestein via 2*len-1-padded pow2 passes)");
         return nullptr;
     }
     fsdr_filter* f = create_common(K_FFT_CF32);
@@ -2322,18 +2385,75 @@ extern "C" fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse,
     f->fft_shift = fft_shift;
     f->norm = normalize ? *normalize : 0.f;
     f->n_taps = len; /* length() = min_items = len (fft.rs:106-109) */
-    /* twiddle table W[k] = e^{-2πik/len}, k < len, computed in f64
-     * (radix-4 needs indices up to 3(len/4-1)) */
-    std::vector<float2> tw(len);
-    for (size_t k = 0; k < len; k++) {
-        double a = -2.0 * M_PI * (double)k / (double)len;
+    size_t tlen = len;
+    if (!pow2) {
+        size_t M = 4;
+        while (M < 2 * len - 1) M <<= 1;
+        f->fft_m = M;
+        tlen = M; /* twiddles for the M-point passes */
+        /* chirp w(k) = e^{sigma*pi*i*k^2/len}, sigma = -1 fwd / +1 inv;
+         * k^2 reduced mod 2*len in integers before the f64 angle */
+        const double sg = inverse ? 1.0 : -1.0;
+        std::vector<float2> w(len);
+        std::vector<double> wr(len), wi(len);
+        for (size_t k = 0; k < len; k++) {
+            unsigned long long r = (k * k) % (2 * len);
+            double a = sg * M_PI * (double)r / (double)len;
+            wr[k] = cos(a);
+            wi[k] = sin(a);
+            w[k] = make_float2((float)wr[k], (float)wi[k]);
+        }
+        /* V_seq = wrapped conj(w); B = forward DFT_M(V_seq) in f64 */
+        std::vector<double> vr(M, 0.0), vi(M, 0.0);
+        for (size_t j = 0; j < len; j++) {
+            vr[j] = wr[j];
+            vi[j] = -wi[j];
+            if (j) {
+                vr[M - j] = wr[j];
+                vi[M - j] = -wi[j];
+            }
+        }
+        std::vector<double> cm(M), sm(M);
+        for (size_t k = 0; k < M; k++) {
+            double a = -2.0 * M_PI * (double)k / (double)M;
+            cm[k] = cos(a);
+            sm[k] = sin(a);
+        }
+        std::vector<float2> B(M);
+        for (size_t k = 0; k < M; k++) {
+            double sre = 0.0, sim = 0.0;
+            for (size_t j2 = 0; j2 < M; j2++) {
+                if (vr[j2] == 0.0 && vi[j2] == 0.0) continue;
+                size_t idx = (k * j2) % M;
+                double c = cm[idx], s = sm[idx];
+                sre += vr[j2] * c - vi[j2] * s;
+                sim += vr[j2] * s + vi[j2] * c;
+            }
+            B[k] = make_float2((float)sre, (float)sim);
+        }
+        if (hipMalloc(&f->d_chirp, len * sizeof(float2)) != hipSuccess ||
+            hipMemcpy(f->d_chirp, w.data(), len * sizeof(float2),
+                      hipMemcpyHostToDevice) != hipSuccess ||
+            hipMalloc(&f->d_B, M * sizeof(float2)) != hipSuccess ||
+            hipMemcpy(f->d_B, B.data(), M * sizeof(float2),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_err("bluestein table upload failed");
+            fsdr_filter_destroy(f);
+            return nullptr;
+        }
+    }
+    /* twiddle table W[k] = e^{-2πik/tlen}, k < tlen, computed in f64
+     * (radix-4 needs indices up to 3(tlen/4-1)) */
+    std::vector<float2> tw(tlen);
+    for (size_t k = 0; k < tlen; k++) {
+        double a = -2.0 * M_PI * (double)k / (double)tlen;
         tw[k] = make_float2((float)cos(a), (float)sin(a));
     }
     if (hipMalloc(&f->d_twid, tw.size() * sizeof(float2)) != hipSuccess ||
         hipMemcpy(f->d_twid, tw.data(), tw.size() * sizeof(float2),
                   hipMemcpyHostToDevice) != hipSuccess) {
         set_err("twiddle upload failed");
-        delete f;
+        fsdr_filter_destroy(f);
         return nullptr;
     }
     return f;
@@ -2484,6 +2604,8 @@ extern "C" void fsdr_filter_destroy(fsdr_filter* f) {
     if (f->d_mtaps) (void)hipFree(f->d_mtaps);
     if (f->d_mtaps32) (void)hipFree(f->d_mtaps32);
     if (f->d_twid) (void)hipFree(f->d_twid);
+    if (f->d_chirp) (void)hipFree(f->d_chirp);
+    if (f->d_B) (void)hipFree(f->d_B);
     if (f->d_in) (void)hipFree(f->d_in);
     if (f->d_out) (void)hipFree(f->d_out);
     if (f->d_scratch) (void)hipFree(f->d_scratch);
@@ -2662,11 +2784,10 @@ static int launch_decim_cf32(fsdr_filter* f, const void* d_in, void* d_out,
     return FSDR_OK;
 }
 
-static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
-                      size_t frames, hipStream_t st,
-                      float* d_mag = nullptr) {
-    if (frames == 0) return FSDR_OK;
-    int n = (int)f->fft_len;
+static int launch_stockham(const float2* d_in, float2* d_out,
+                           float* d_mag, const float2* d_twid, int n,
+                           size_t frames, int inverse, int fft_shift,
+                           float norm, hipStream_t st) {
     int log2n = 0;
     while ((1 << log2n) < n) log2n++;
     int fpb_base = 1024;
@@ -2678,11 +2799,52 @@ static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
     long long blocks = ((long long)frames + fpb - 1) / fpb;
     int grid = (int)std::min<long long>(blocks, 256 * 16);
     hipLaunchKernelGGL(k_fft_stockham, dim3(grid), dim3(256), lds, st,
-                       (const float2*)d_in, (float2*)d_out, d_mag,
-                       f->d_twid, n, log2n, fpb, f->inverse, f->fft_shift,
-                       f->norm, (long long)frames);
+                       d_in, d_out, d_mag, d_twid, n, log2n, fpb, inverse,
+                       fft_shift, norm, (long long)frames);
     HIP_TRY(hipGetLastError());
     return FSDR_OK;
+}
+
+static int launch_fft(fsdr_filter* f, const void* d_in, void* d_out,
+                      size_t frames, hipStream_t st,
+                      float* d_mag = nullptr) {
+    if (frames == 0) return FSDR_OK;
+    int n = (int)f->fft_len;
+    if (f->fft_m) { /* Bluestein: pre -> FFT_M -> *B -> IFFT_M -> post */
+        int M = (int)f->fft_m;
+        int rc = ensure_dev(&f->d_scratch, &f->d_scratch_bytes,
+                            2 * frames * (size_t)M * sizeof(float2));
+        if (rc) return rc;
+        float2* s1 = (float2*)f->d_scratch;
+        float2* s2 = s1 + frames * (size_t)M;
+        long long tM = (long long)frames * M;
+        hipLaunchKernelGGL(k_bluestein_pre, dim3(grid_for(tM, 256)),
+                           dim3(256), 0, st, (const float2*)d_in, s1,
+                           f->d_chirp, n, M, (long long)frames,
+                           f->inverse && f->fft_shift);
+        HIP_TRY(hipGetLastError());
+        rc = launch_stockham(s1, s2, nullptr, f->d_twid, M, frames, 0, 0,
+                             0.f, st);
+        if (rc) return rc;
+        hipLaunchKernelGGL(k_bluestein_mul, dim3(grid_for(tM, 256)),
+                           dim3(256), 0, st, s2, f->d_B, M,
+                           (long long)frames);
+        HIP_TRY(hipGetLastError());
+        rc = launch_stockham(s2, s1, nullptr, f->d_twid, M, frames, 1, 0,
+                             0.f, st);
+        if (rc) return rc;
+        float scale = (1.0f / (float)M) * (f->norm != 0.f ? f->norm : 1.f);
+        hipLaunchKernelGGL(k_bluestein_post,
+                           dim3(grid_for((long long)frames * n, 256)),
+                           dim3(256), 0, st, s1, (float2*)d_out, d_mag,
+                           f->d_chirp, n, M, (long long)frames,
+                           (!f->inverse) && f->fft_shift, scale);
+        HIP_TRY(hipGetLastError());
+        return FSDR_OK;
+    }
+    return launch_stockham((const float2*)d_in, (float2*)d_out, d_mag,
+                           f->d_twid, n, frames, f->inverse, f->fft_shift,
+                           f->norm, st);
 }
 
 /* Bulk batch FFT: the GPU-native many-frames path (what the chain uses

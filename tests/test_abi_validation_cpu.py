@@ -48,6 +48,8 @@ def test_pfb_constraints(fsdr):
 
 
 def test_fft_len_gate(fsdr):
-    assert not fsdr.lib().fsdr_fft_cf32_create(3, 0, 0, None)
+    # pow2 lengths: [4,4096]; any other length: [2,2048] via Bluestein
+    assert not fsdr.lib().fsdr_fft_cf32_create(1, 0, 0, None)
     assert not fsdr.lib().fsdr_fft_cf32_create(8192, 0, 0, None)
-    assert "power of two" in _err(fsdr)
+    assert not fsdr.lib().fsdr_fft_cf32_create(3000, 0, 0, None)
+    assert "Bluestein" in _err(fsdr)

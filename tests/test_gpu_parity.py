@@ -254,6 +254,45 @@ def test_fft_parseval(gpu):
         assert abs(e_t - e_f) / e_t < 1e-5
 
 
+@pytest.mark.parametrize("n", [6, 12, 60, 100, 384, 1000, 2000])
+def test_fft_bluestein_parity_vs_oracle(gpu, oracle_lib, n):
+    """Non-pow2 lengths (the reference Fft is generic over rustfft plan
+    lengths, fft.rs:98-103): Bluestein path vs the f64 oracle DFT."""
+    r = rng(4000 + n)
+    frames = 5
+    x = cplx(r, n * frames)
+    got, c, p, s = gpu.Fft(n).filter(x, n * frames)
+    assert (c, p) == (n * frames, n * frames)
+    ref = np.concatenate([
+        oracle_lib.dft_cf32(x[i * n:(i + 1) * n]) for i in range(frames)])
+    rel = np.linalg.norm(got - ref) / np.linalg.norm(ref)
+    assert rel < 2e-4, rel
+
+
+def test_fft_bluestein_inverse_shift_normalize(gpu, oracle_lib):
+    r = rng(4100)
+    n = 100
+    x = cplx(r, n * 3)
+    for kw in [dict(inverse=True), dict(fft_shift=True),
+               dict(inverse=True, fft_shift=True),
+               dict(normalize=1.0 / n)]:
+        got, c, p, s = gpu.Fft(n, **kw).filter(x, n * 3)
+        ref, m = oracle_lib.fft_block(n, x, n * 3, **kw)
+        assert p == m
+        rel = np.linalg.norm(got - ref) / max(np.linalg.norm(ref), 1e-30)
+        assert rel < 2e-4, (kw, rel)
+
+
+def test_fft_bluestein_roundtrip(gpu):
+    r = rng(4200)
+    n = 60
+    x = cplx(r, n * 4)
+    X, _, _, _ = gpu.Fft(n).filter(x, x.size)
+    back, _, _, _ = gpu.Fft(n, inverse=True, normalize=1.0 / n).filter(
+        X, X.size)
+    assert_close(back, x, 1e-4)
+
+
 # ---------------- element-wise + synthetic source ----------------------
 
 def test_mag2_parity(gpu, oracle_lib):
