@@ -98,8 +98,11 @@ fsdr_filter* fsdr_decim_fir_cf32_create(size_t decimation,
  * multiple of interp (polyphase_resampling_fir.rs:54-56 assert). */
 fsdr_filter* fsdr_resamp_cf32_create(size_t interp, size_t decim,
                                      const float* taps, size_t n_taps);
-/* Fft block — len must be a power of two in [16, 4096]; inverse/fft_shift
- * flags and optional normalize factor as src/blocks/fft.rs:92-121. */
+/* Fft block — pow2 lengths in [4,4096] run the radix-4 Stockham kernel;
+ * ANY other length in [2,2048] runs via Bluestein (two pow2 M>=2len-1
+ * passes; the reference block is generic over rustfft plan lengths,
+ * fft.rs:98-103). inverse/fft_shift flags and optional normalize factor
+ * as src/blocks/fft.rs:92-121. */
 fsdr_filter* fsdr_fft_cf32_create(size_t len, int inverse, int fft_shift,
                                   const float* normalize);
 /* Apply |x|^2 (Complex32 -> f32) — the spectrum mag^2 map. */
