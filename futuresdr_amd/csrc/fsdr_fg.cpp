@@ -16,8 +16,12 @@
  *    with one worker), which is perf/fir's default (perf/fir/fir.rs:77-80).
  *
  * Data stays resident in HBM between blocks; only sources/sinks touch host
- * memory. This is the harness the fg-level tests and INTEGRATION.md refer
- * to; the fused fsdr_chain_* path is the production pipeline.
+ * memory. All data-plane work (kernels, D2D compaction, source fills) is
+ * ENQUEUED asynchronously on one HIP stream — the host actor loop's
+ * bookkeeping is pure status math, so it runs ahead of the device and the
+ * only synchronization points are capture sinks and run() completion.
+ * This is the harness the fg-level tests and INTEGRATION.md refer to; the
+ * fused fsdr_chain_* path is the production pipeline.
  */
 #include <hip/hip_runtime.h>
 
@@ -65,6 +69,12 @@ struct fsdr_fg {
     std::vector<Edge> edges;
     size_t default_cap = 1 << 18; /* items per stream buffer */
     bool dead = false;
+    /* all data-plane work is ENQUEUED on this stream; the actor loop's
+     * bookkeeping is host-side status math (no readbacks), so the
+     * round-robin runs ahead of the GPU and the only syncs are capture
+     * sinks and run() end — the async analogue of the reference's
+     * per-block tasks overlapping on one queue. */
+    hipStream_t stream = nullptr;
 };
 
 static bool fg_have_gpu() {
@@ -74,11 +84,20 @@ static bool fg_have_gpu() {
 
 extern "C" fsdr_fg* fsdr_fg_create(void) {
     if (!fg_have_gpu()) return nullptr; /* product path: no CPU fallback */
-    return new fsdr_fg();
+    fsdr_fg* fg = new fsdr_fg();
+    if (hipStreamCreate(&fg->stream) != hipSuccess) {
+        delete fg;
+        return nullptr;
+    }
+    return fg;
 }
 
 extern "C" void fsdr_fg_destroy(fsdr_fg* fg) {
     if (!fg) return;
+    if (fg->stream) {
+        (void)hipStreamSynchronize(fg->stream);
+        (void)hipStreamDestroy(fg->stream);
+    }
     for (auto& e : fg->edges)
         if (e.dev) (void)hipFree(e.dev);
     delete fg;
@@ -155,7 +174,7 @@ extern "C" int fsdr_fg_stream(fsdr_fg* fg, int src, int dst) {
 /* compaction = the slab reserved-prefix tail copy (slab.rs:369-399):
  * move [r - reserved, w) to the buffer start so the writer regains space
  * while the reader keeps its history. */
-static int edge_compact(Edge& e) {
+static int edge_compact(Edge& e, hipStream_t st) {
     size_t keep_from = e.r;
     size_t keep = e.w - keep_from;
     if (keep_from == 0) return FSDR_OK; /* nothing to gain */
@@ -165,20 +184,21 @@ static int edge_compact(Edge& e) {
          * and we only compact when the writer is starved; use a bounce
          * via memcpyDtoD which requires non-overlap — guard it. */
         if (keep_from >= keep) {
-            if (hipMemcpy((char*)e.dev,
-                          (char*)e.dev + keep_from * e.item_bytes,
-                          keep * e.item_bytes,
-                          hipMemcpyDeviceToDevice) != hipSuccess)
+            if (hipMemcpyAsync((char*)e.dev,
+                               (char*)e.dev + keep_from * e.item_bytes,
+                               keep * e.item_bytes,
+                               hipMemcpyDeviceToDevice, st) != hipSuccess)
                 return FSDR_ERR_HIP;
         } else {
-            /* overlapping: chunked forward copy */
+            /* overlapping: chunked forward copy (same stream = ordered) */
             size_t done = 0;
             while (done < keep) {
                 size_t c = keep_from < keep - done ? keep_from : keep - done;
-                if (hipMemcpy((char*)e.dev + done * e.item_bytes,
-                              (char*)e.dev + (keep_from + done) * e.item_bytes,
-                              c * e.item_bytes,
-                              hipMemcpyDeviceToDevice) != hipSuccess)
+                if (hipMemcpyAsync(
+                        (char*)e.dev + done * e.item_bytes,
+                        (char*)e.dev + (keep_from + done) * e.item_bytes,
+                        c * e.item_bytes, hipMemcpyDeviceToDevice,
+                        st) != hipSuccess)
                     return FSDR_ERR_HIP;
                 done += c;
             }
@@ -198,12 +218,13 @@ static long long block_work(fsdr_fg* fg, Block& b) {
         case B_NULL_SRC: { /* null_source.rs:53-66: zero-fill all space */
             if (!oe) return 0;
             if (oe->writable() == 0) {
-                if (edge_compact(*oe) != FSDR_OK) return -1;
+                if (edge_compact(*oe, fg->stream) != FSDR_OK) return -1;
             }
             size_t n = oe->writable();
             if (n == 0) return 0;
-            if (hipMemset((char*)oe->dev + oe->w * oe->item_bytes, 0,
-                          n * oe->item_bytes) != hipSuccess)
+            if (hipMemsetAsync((char*)oe->dev + oe->w * oe->item_bytes, 0,
+                               n * oe->item_bytes,
+                               fg->stream) != hipSuccess)
                 return -1;
             oe->w += n;
             return (long long)n;
@@ -216,13 +237,14 @@ static long long block_work(fsdr_fg* fg, Block& b) {
                 oe->writer_finished = true;
                 return 0;
             }
-            if (oe->writable() == 0 && edge_compact(*oe) != FSDR_OK)
+            if (oe->writable() == 0 && edge_compact(*oe, fg->stream) != FSDR_OK)
                 return -1;
             size_t n = oe->writable() < left ? oe->writable() : left;
             if (n == 0) return 0;
-            if (hipMemcpy((char*)oe->dev + oe->w * oe->item_bytes,
-                          b.vec.data() + b.vec_pos, n * oe->item_bytes,
-                          hipMemcpyHostToDevice) != hipSuccess)
+            if (hipMemcpyAsync((char*)oe->dev + oe->w * oe->item_bytes,
+                               b.vec.data() + b.vec_pos,
+                               n * oe->item_bytes, hipMemcpyHostToDevice,
+                               fg->stream) != hipSuccess)
                 return -1;
             oe->w += n;
             b.vec_pos += n * oe->item_bytes;
@@ -234,14 +256,15 @@ static long long block_work(fsdr_fg* fg, Block& b) {
             size_t n = avail;
             if ((unsigned long long)n > b.head_n) n = (size_t)b.head_n;
             if (oe->writable() < n) {
-                if (edge_compact(*oe) != FSDR_OK) return -1;
+                if (edge_compact(*oe, fg->stream) != FSDR_OK) return -1;
                 if (oe->writable() < n) n = oe->writable();
             }
             if (n > 0) {
-                if (hipMemcpy((char*)oe->dev + oe->w * oe->item_bytes,
-                              (char*)ie->dev + ie->r * ie->item_bytes,
-                              n * ie->item_bytes,
-                              hipMemcpyDeviceToDevice) != hipSuccess)
+                if (hipMemcpyAsync((char*)oe->dev + oe->w * oe->item_bytes,
+                                   (char*)ie->dev + ie->r * ie->item_bytes,
+                                   n * ie->item_bytes,
+                                   hipMemcpyDeviceToDevice,
+                                   fg->stream) != hipSuccess)
                     return -1;
                 ie->r += n;
                 oe->w += n;
@@ -261,17 +284,18 @@ static long long block_work(fsdr_fg* fg, Block& b) {
              * is the history the next call sees. */
             size_t n_in = ie->readable();
             if (oe->writable() < fg->default_cap / 2 &&
-                edge_compact(*oe) != FSDR_OK)
+                edge_compact(*oe, fg->stream) != FSDR_OK)
                 return -1;
             size_t n_out = oe->writable();
             fsdr_filter_result r;
             int rc = fsdr_filter_dev(
                 b.filter, (char*)ie->dev + ie->r * ie->item_bytes, n_in,
-                (char*)oe->dev + oe->w * oe->item_bytes, n_out, nullptr, &r);
+                (char*)oe->dev + oe->w * oe->item_bytes, n_out, fg->stream,
+                &r);
             if (rc != FSDR_OK) return -1;
             ie->r += r.consumed;
             oe->w += r.produced;
-            if (ie->r > 0) (void)edge_compact(*ie);
+            if (ie->r > 0) (void)edge_compact(*ie, fg->stream);
             /* fir.rs:89-91 / wrapped_kernel finish propagation: upstream
              * finished and not output-limited -> done (the unconsumable
              * tail < min_items is dropped, like the reference) */
@@ -287,7 +311,7 @@ static long long block_work(fsdr_fg* fg, Block& b) {
             size_t n = ie->readable();
             b.n_received += n;
             ie->r += n;
-            if (n > 0) (void)edge_compact(*ie);
+            if (n > 0) (void)edge_compact(*ie, fg->stream);
             if (ie->writer_finished && ie->readable() == 0)
                 b.finished = true;
             return (long long)n;
@@ -298,6 +322,10 @@ static long long block_work(fsdr_fg* fg, Block& b) {
             if (n > 0) {
                 size_t old = b.vec.size();
                 b.vec.resize(old + n * ie->item_bytes);
+                /* capture sink: drain enqueued work, then a blocking
+                 * copy into the (reallocatable) host vector */
+                if (hipStreamSynchronize(fg->stream) != hipSuccess)
+                    return -1;
                 if (hipMemcpy(b.vec.data() + old,
                               (char*)ie->dev + ie->r * ie->item_bytes,
                               n * ie->item_bytes,
@@ -305,7 +333,7 @@ static long long block_work(fsdr_fg* fg, Block& b) {
                     return -1;
                 ie->r += n;
                 b.n_received += n;
-                (void)edge_compact(*ie);
+                (void)edge_compact(*ie, fg->stream);
             }
             if (ie->writer_finished && ie->readable() == 0)
                 b.finished = true;
@@ -350,7 +378,7 @@ extern "C" int fsdr_fg_run(fsdr_fg* fg) {
             return FSDR_ERR_INVALID;
         }
     }
-    (void)hipDeviceSynchronize();
+    (void)hipStreamSynchronize(fg->stream);
     return FSDR_OK;
 }
 
