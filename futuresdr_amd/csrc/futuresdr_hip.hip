@@ -1469,6 +1469,68 @@ __global__ __launch_bounds__(256) void k_xlating_decim_ccf32(
     }
 }
 
+/* LDS-tiled XlatingFir (same output math as k_xlating_decim_ccf32): a
+ * tile of XT_TILE outputs stages its whole input span into padded SoA
+ * planes once; complex bpf taps in LDS. Fallback to the naive kernel
+ * when the span exceeds the LDS budget (host decides). */
+#define XT_BLOCK 256
+#define XT_R 2
+#define XT_TILE (XT_BLOCK * XT_R)
+
+__global__ __launch_bounds__(XT_BLOCK) void k_xlating_decim_tiled_ccf32(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float2* __restrict__ taps /* reversed bpf taps */, int n_taps,
+    int decim, long long n_out, long long n_in_valid, double theta,
+    float p0r, float p0i, int span) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* s_re = (float*)smem;
+    float* s_im = s_re + plane_floats((unsigned)span);
+    float* s_hr = s_im + plane_floats((unsigned)span);
+    float* s_hi = s_hr + n_taps;
+    for (int i = threadIdx.x; i < n_taps; i += XT_BLOCK) {
+        s_hr[i] = taps[i].x;
+        s_hi[i] = taps[i].y;
+    }
+    const long long tiles = (n_out + XT_TILE - 1) / XT_TILE;
+    for (long long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
+        const long long out_base = tile * XT_TILE;
+        const long long in_base = decim - 1 + out_base * decim;
+        __syncthreads();
+        for (int i = threadIdx.x; i < span; i += XT_BLOCK) {
+            long long g = in_base + i;
+            float2 v = (g < n_in_valid) ? in[g] : make_float2(0.f, 0.f);
+            s_re[lds_pad((unsigned)i)] = v.x;
+            s_im[lds_pad((unsigned)i)] = v.y;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int r = 0; r < XT_R; r++) {
+            long long k = out_base + threadIdx.x + r * XT_BLOCK;
+            if (k >= n_out) break;
+            unsigned rel = (unsigned)((k - out_base) * decim);
+            float sre = 0.f, sim = 0.f;
+#pragma unroll 4
+            for (int t = 0; t < n_taps; t++) {
+                unsigned a = lds_pad(rel + (unsigned)t);
+                float xr = s_re[a], xi = s_im[a];
+                float hr = s_hr[t], hi = s_hi[t];
+                sre = fmaf(xr, hr, sre);
+                sre = fmaf(-xi, hi, sre);
+                sim = fmaf(xr, hi, sim);
+                sim = fmaf(xi, hr, sim);
+            }
+            /* f64 angle + mod-2pi reduction (see k_rotator) */
+            double a = theta * (double)(k + 1);
+            a -= 6.283185307179586 * floor(a * 0.15915494309189535);
+            float s, cth;
+            __sincosf((float)a, &s, &cth);
+            float pr = cth * p0r - s * p0i;
+            float pi = cth * p0i + s * p0r;
+            out[k] = make_float2(sre * pr - sim * pi, sre * pi + sim * pr);
+        }
+    }
+}
+
 /* ================= WLAN sync-short helpers ============================ *
  * examples/wlan/src/bin/rx.rs:73-96 autocorrelation chain pieces:
  * a*conj(b) Combine (:81) and the sliding-SUM MovingAverage
@@ -2961,15 +3023,35 @@ extern "C" int fsdr_filter_dev(fsdr_filter* f, const void* d_in, size_t n_in,
         case K_XLATING: {
             *r = decim_status(f->decim, n_in, f->n_taps, n_out);
             if (r->produced == 0) return FSDR_OK;
-            size_t lds = f->n_taps * sizeof(float2);
-            hipLaunchKernelGGL(k_xlating_decim_ccf32,
-                               dim3(grid_for((long long)r->produced, 256)),
-                               dim3(256), lds, st, (const float2*)d_in,
-                               (float2*)d_out, (const float2*)f->d_taps,
-                               (int)f->n_taps, (long long)f->decim,
-                               (long long)r->produced, (long long)n_in,
-                               f->theta, f->rot_re, f->rot_im);
-            HIP_TRY(hipGetLastError());
+            long long span = (long long)(XT_TILE - 1) * f->decim +
+                             f->n_taps + 2;
+            size_t lds_t = (2 * (size_t)plane_floats((unsigned)span) +
+                            2 * f->n_taps) * sizeof(float);
+            const char* xt = getenv("FSDR_XLATING_TILED");
+            if (lds_t <= 64 * 1024 && (!xt || atoi(xt) != 0)) {
+                long long tiles =
+                    ((long long)r->produced + XT_TILE - 1) / XT_TILE;
+                int grid = (int)std::min<long long>(tiles, 256 * 64);
+                hipLaunchKernelGGL(k_xlating_decim_tiled_ccf32,
+                                   dim3(grid), dim3(XT_BLOCK), lds_t, st,
+                                   (const float2*)d_in, (float2*)d_out,
+                                   (const float2*)f->d_taps,
+                                   (int)f->n_taps, (int)f->decim,
+                                   (long long)r->produced, (long long)n_in,
+                                   (double)f->theta, f->rot_re, f->rot_im,
+                                   (int)span);
+                HIP_TRY(hipGetLastError());
+            } else {
+                size_t lds = f->n_taps * sizeof(float2);
+                hipLaunchKernelGGL(
+                    k_xlating_decim_ccf32,
+                    dim3(grid_for((long long)r->produced, 256)), dim3(256),
+                    lds, st, (const float2*)d_in, (float2*)d_out,
+                    (const float2*)f->d_taps, (int)f->n_taps,
+                    (long long)f->decim, (long long)r->produced,
+                    (long long)n_in, f->theta, f->rot_re, f->rot_im);
+                HIP_TRY(hipGetLastError());
+            }
             { /* advance the rotator phase (closed form, f64) */
                 double a = (double)f->theta * (double)r->produced;
                 double cs = cos(a), sn = sin(a);
