@@ -691,7 +691,22 @@ size_t oracle_chain_cf32(const float* taps1, size_t n_taps1,
                 }
                 fre[k] = sre; fim[k] = sim;
             }
-            fft_f32_pow2((int)fft_len, fre, fim);
+            if ((fft_len & (fft_len - 1)) == 0) {
+                fft_f32_pow2((int)fft_len, fre, fim);
+            } else { /* non-pow2 frames: exact f64 DFT (test infra) */
+                double* dr = (double*)malloc(2 * fft_len * sizeof(double));
+                double* di = dr + fft_len;
+                for (size_t q = 0; q < fft_len; q++) {
+                    dr[q] = fre[q];
+                    di[q] = fim[q];
+                }
+                dft_f64((int)fft_len, 0, dr, di);
+                for (size_t q = 0; q < fft_len; q++) {
+                    fre[q] = (float)dr[q];
+                    fim[q] = (float)di[q];
+                }
+                free(dr);
+            }
             if (out_spectra) {
                 for (size_t k = 0; k < fft_len; k++) {
                     out_spectra[y2_base + k].re = fre[k];
@@ -776,7 +791,22 @@ size_t oracle_chain_cf32_fast(const float* taps1, size_t n_taps1,
                 fre[k] = sre;
                 fim[k] = sim;
             }
-            fft_f32_pow2((int)fft_len, fre, fim);
+            if ((fft_len & (fft_len - 1)) == 0) {
+                fft_f32_pow2((int)fft_len, fre, fim);
+            } else { /* non-pow2 frames: exact f64 DFT (test infra) */
+                double* dr = (double*)malloc(2 * fft_len * sizeof(double));
+                double* di = dr + fft_len;
+                for (size_t q = 0; q < fft_len; q++) {
+                    dr[q] = fre[q];
+                    di[q] = fim[q];
+                }
+                dft_f64((int)fft_len, 0, dr, di);
+                for (size_t q = 0; q < fft_len; q++) {
+                    fre[q] = (float)dr[q];
+                    fim[q] = (float)di[q];
+                }
+                free(dr);
+            }
             if (out_spectra) {
                 for (size_t k = 0; k < fft_len; k++) {
                     out_spectra[y2_base + k].re = fre[k];

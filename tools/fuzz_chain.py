@@ -22,7 +22,8 @@ def main(n_cases=40):
     for case in range(n_cases):
         nt1 = int(r.integers(2, 400))
         nt2 = int(r.integers(2, 400))
-        fft_len = int(r.choice([64, 256, 512, 1024, 2048]))
+        fft_len = int(r.choice([64, 100, 256, 384, 512, 1000, 1024,
+                                2048]))  # non-pow2 -> Bluestein split path
         n_in = int(r.integers(nt1 + nt2 + 4 * fft_len, 300_000))
         t1 = r.uniform(-1, 1, nt1).astype(np.float32)
         t2 = r.uniform(-1, 1, nt2).astype(np.float32)
