@@ -987,6 +987,154 @@ __global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
     }
 }
 
+/* All-phase staging variant (FSDR_CHAIN_ALLPHASE=1): all 4 phase planes
+ * resident at once (8 planes, ~36 KB LDS -> 4 blocks/CU instead of 6),
+ * ONE staging write + barrier per tile and a contiguous 160-MFMA run —
+ * trades occupancy for fewer barrier convoys. Experiment for the
+ * phase-structure bound documented in profiles/pmc_sq_r02.txt. */
+template <int KKD>
+__global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ap_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv /* [4][KKD], rtv[v][u] = rt[4u+v] */,
+    long long n_out, long long n_in_valid,
+    const float2* __restrict__ twid /* fft_len-entry forward table */,
+    float* __restrict__ mag_out /* nullable |X|^2 */,
+    int fft_len) {
+    static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
+    const unsigned elemsP = MDFIR_TILE + KKD + 8;
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    const unsigned SUB = SPm / 4;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* planes = (float*)smem;        /* [8][SPm]: re v0..3, im v0..3 */
+    float* s_rtx = planes + 8u * SPm;    /* [4][KKD+16] */
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+
+    for (int i = tid; i < 4 * (KKD + 16); i += MDFIR_BLOCK) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    __syncthreads();
+
+    const unsigned span = 3 + 4 * elemsP;
+    constexpr int NL4 =
+        (4 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float2 stg[NL4];
+    auto load_all = [&](long long tl) {
+        const long long ib = tl * MDFIR_TILE * 4;
+#pragma unroll
+        for (int j = 0; j < NL4; j++) {
+            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+            unsigned rel = 3 + idx;
+            long long g = ib + rel;
+            stg[j] = (rel < span && g < n_in_valid)
+                         ? in[g] : make_float2(0.f, 0.f);
+        }
+    };
+    auto write_all = [&]() {
+#pragma unroll
+        for (int j = 0; j < NL4; j++) {
+            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+            unsigned v = idx & 3u, u = idx >> 2;
+            if (u < elemsP) {
+                unsigned d = v * SPm + (u & 3u) * SUB + (u >> 2);
+                planes[d] = stg[j].x;
+                planes[4u * SPm + d] = stg[j].y;
+            }
+        }
+    };
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+
+    load_all(blockIdx.x);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MDFIR_TILE;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        write_all();
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_all(tile + gridDim.x); /* under this tile's MFMAs */
+#pragma unroll
+        for (int v = 0; v < 4; v++) {
+            const float* pre = planes + (unsigned)v * SPm + asub;
+            const float* pim = planes + (4u + v) * SPm + asub;
+            float bfrag[KKD / 4];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++)
+                bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int t = 0; t < (KKD / 4) / 4; t++) {
+                float4 ar = *(const float4*)&pre[abase + 4 * t];
+                float4 ai = *(const float4*)&pim[abase + 4 * t];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.x, bfrag[4 * t], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.x, bfrag[4 * t], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
+            }
+#pragma unroll
+            for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
+                float a_re = pre[abase + s];
+                float a_im = pim[abase + s];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
+            }
+            __builtin_amdgcn_s_setprio(0);
+        }
+        __syncthreads(); /* planes dead; reuse as FFT LDS */
+        float2* ping = (float2*)planes;
+        float2* pong = ping + 1024;
+        const unsigned Lm = (unsigned)fft_len - 1u;
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            unsigned pos = wave * 256 + 16 * row + r16;
+            ping[(pos & ~Lm) | fft_swz(pos & Lm)] =
+                make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+        {
+            const int F = 1024 / fft_len;
+            const int tpf = MDFIR_BLOCK / F;
+            const int fl = tid / tpf, tfr = tid - fl * tpf;
+            float2* res = fft_pow2_fwd(ping + (size_t)fl * fft_len,
+                                       pong + (size_t)fl * fft_len, twid,
+                                       fft_len, tfr, tpf) -
+                          (size_t)fl * fft_len;
+            for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+                long long o = out_base + i;
+                if (o < n_out) {
+                    float2 v =
+                        res[((unsigned)i & ~Lm) | fft_swz((unsigned)i & Lm)];
+                    if (out) out[o] = v;
+                    if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
 template <int KKD>
 __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
     const float2* __restrict__ in, float2* __restrict__ out,
@@ -3866,6 +4014,35 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                                (long long)n_in,
                                (const float2*)c->fft->d_twid,
                                (float*)d_mag);
+            HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
+        const char* ap = getenv("FSDR_CHAIN_ALLPHASE");
+        if (ap && atoi(ap) != 0) {
+            unsigned SPm = (elemsP + 31u) & ~31u;
+            size_t lds_ap =
+                (8 * (size_t)SPm + 4 * ((size_t)KK + 16)) * sizeof(float);
+#define CHAIN_AP_CASE(KV)                                                 \
+    case KV:                                                              \
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_ap_tpl<KV>), \
+                           dim3(grid), dim3(MDFIR_BLOCK), lds_ap, st,     \
+                           (const float2*)d_in, spec_dst,                 \
+                           c->fused->d_mtaps, (long long)prod,            \
+                           (long long)n_in,                               \
+                           (const float2*)c->fft->d_twid, (float*)d_mag,  \
+                           (int)L);                                       \
+        break;
+            switch (KK) {
+                CHAIN_AP_CASE(20)
+                CHAIN_AP_CASE(32)
+                CHAIN_AP_CASE(48)
+                CHAIN_AP_CASE(80)
+                CHAIN_AP_CASE(144)
+                default:
+                    set_err("bad chain mfma K");
+                    return FSDR_ERR_INVALID;
+            }
+#undef CHAIN_AP_CASE
             HIP_TRY(hipGetLastError());
             return FSDR_OK;
         }
