@@ -1303,6 +1303,109 @@ __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
     }
 }
 
+/* MFMA-loop microbenchmark (tools/mfma_ubench.py): the decim kernel's
+ * inner loop on LDS staged ONCE — no per-tile staging, no FFT, no
+ * barriers in the hot loop. Measures the intrinsic ceiling of the
+ * b128-fed 16x16x4-f32 loop at this occupancy. */
+template <int KKD>
+__global__ __launch_bounds__(MDFIR_BLOCK) void k_mfma_ubench_tpl(
+    const float* __restrict__ rtv, float2* __restrict__ out, int iters) {
+    const unsigned elemsP = MDFIR_TILE + KKD + 8;
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    const unsigned SUB = SPm / 4;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* planes = (float*)smem;
+    float* s_rtx = planes + 4u * SPm;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+    for (int i = tid; i < 4 * (KKD + 16); i += MDFIR_BLOCK) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    for (int i = tid; i < (int)(4 * SPm); i += MDFIR_BLOCK)
+        planes[i] = (float)(i & 255) * 0.001f;
+    __syncthreads();
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+    v4f cre = {0.f, 0.f, 0.f, 0.f};
+    v4f cim = {0.f, 0.f, 0.f, 0.f};
+    for (int it = 0; it < iters; it++) {
+#pragma unroll
+        for (int vloc = 0; vloc < 2; vloc++) {
+            const float* pre = planes + (unsigned)vloc * SPm + asub;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm + asub;
+            const int v = 2 * ((it ^ vloc) & 1) + vloc;
+            float bfrag[KKD / 4];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++)
+                bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+#pragma unroll
+            for (int t = 0; t < (KKD / 4) / 4; t++) {
+                float4 ar = *(const float4*)&pre[abase + 4 * t];
+                float4 ai = *(const float4*)&pim[abase + 4 * t];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.x, bfrag[4 * t], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.x, bfrag[4 * t], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
+            }
+        }
+    }
+    long long o = (long long)blockIdx.x * MDFIR_BLOCK + tid;
+    out[o] = make_float2(cre[0] + cre[1] + cre[2] + cre[3],
+                         cim[0] + cim[1] + cim[2] + cim[3]);
+}
+
+extern "C" int fsdr_mfma_ubench(int grid, int iters, double* tflops,
+                                void* stream) {
+    REQUIRE_GPU();
+    const int KKD = 80;
+    unsigned elemsP = MDFIR_TILE + KKD + 8;
+    unsigned SPm = (elemsP + 31u) & ~31u;
+    size_t lds = (4 * (size_t)SPm + 4 * (KKD + 16)) * sizeof(float);
+    float* d_taps = nullptr;
+    float2* d_out = nullptr;
+    HIP_TRY(hipMalloc(&d_taps, 4 * KKD * sizeof(float)));
+    HIP_TRY(hipMemset(d_taps, 1, 4 * KKD * sizeof(float)));
+    HIP_TRY(hipMalloc(&d_out, (size_t)grid * MDFIR_BLOCK * sizeof(float2)));
+    hipStream_t st = (hipStream_t)stream;
+    hipEvent_t e0, e1;
+    HIP_TRY(hipEventCreate(&e0));
+    HIP_TRY(hipEventCreate(&e1));
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(k_mfma_ubench_tpl<KKD>), dim3(grid),
+                       dim3(MDFIR_BLOCK), lds, st, d_taps, d_out, iters);
+    HIP_TRY(hipEventRecord(e0, st));
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(k_mfma_ubench_tpl<KKD>), dim3(grid),
+                       dim3(MDFIR_BLOCK), lds, st, d_taps, d_out, iters);
+    HIP_TRY(hipEventRecord(e1, st));
+    HIP_TRY(hipEventSynchronize(e1));
+    float ms = 0.f;
+    HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
+    /* flops: grid blocks x 4 waves x iters x 2 vloc x KKD/4 steps x 2
+     * (re+im) MFMAs x 2048 flops */
+    double fl = (double)grid * 4 * iters * 2 * (KKD / 4) * 2 * 2048.0;
+    *tflops = fl / (ms * 1e-3) / 1e12;
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    (void)hipFree(d_taps);
+    (void)hipFree(d_out);
+    return FSDR_OK;
+}
+
 /* ---- Phase-split decimating FIR, D=4, compile-time taps -------------- *
  * Same math as decimating_fir.rs:80-95 (D=4): y[k] = sum_t x[3+4k+t] *
  * h[T-1-t]. Decompose t = 4u+v: y[k] = sum_v sum_u P_v[k+u] * rt[4u+v]

@@ -214,6 +214,10 @@ size_t fsdr_wlan_sync_long_run(fsdr_wlan_rx* rx, const fsdr_cf32* in,
                                size_t* frame_off, float* frame_freq,
                                size_t frame_cap, size_t* num_frames);
 
+/* MFMA-loop microbenchmark (diagnostics; tools/mfma_ubench.py): times
+ * the chain kernel's inner MFMA loop on LDS staged once. */
+int fsdr_mfma_ubench(int grid, int iters, double* tflops, void* stream);
+
 /* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
 int fsdr_dev_alloc(void** d_ptr, size_t bytes);
 int fsdr_dev_free(void* d_ptr);
