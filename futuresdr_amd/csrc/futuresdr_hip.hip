@@ -821,9 +821,18 @@ __global__ __launch_bounds__(MDFIR_BLOCK, MINWG) void k_decim4_fft_mfma_tpl(
     long long n_out, long long n_in_valid,
     const float2* __restrict__ twid /* fft_len-entry forward table */,
     float* __restrict__ mag_out /* nullable |X|^2 */,
-    int fft_len /* pow2 64..1024; tile = 1024/fft_len frames */) {
+    int fft_len /* pow2 64..1024; tile = 1024/fft_len frames */,
+    int stagger /* s_sleep(7) loops per co-residency slot: desyncs the
+                   per-tile phases of co-resident blocks so one block's
+                   staging/FFT interval overlaps the others' MFMA runs
+                   (the bare MFMA loop reaches 97% of peak —
+                   tools/mfma_ubench.py — so pipe idle = phase lock) */) {
     static_assert(MDFIR_TILE == 1024, "tile == 1024 decimated outputs");
     static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
+    if (stagger > 0) {
+        int loops = (int)(blockIdx.x & 7u) * stagger;
+        for (int i = 0; i < loops; i++) __builtin_amdgcn_s_sleep(7);
+    }
     const unsigned elemsP = MDFIR_TILE + KKD + 8;     /* per phase plane */
     const unsigned SPm = (elemsP + 31u) & ~31u;
     /* each phase plane is split into 4 sub-planes by element%4 so a
@@ -4158,10 +4167,13 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                 dim3(grid), dim3(MDFIR_BLOCK), lds, st,
                 (const float2*)d_in, spec_dst, c->fused->d_mtaps,
                 (long long)prod, (long long)n_in,
-                (const float2*)c->fft->d_twid, (float*)d_mag, (int)L);
+                (const float2*)c->fft->d_twid, (float*)d_mag, (int)L, 0);
             HIP_TRY(hipGetLastError());
             return FSDR_OK;
         }
+        int stagger = 0;
+        if (const char* sg = getenv("FSDR_CHAIN_STAGGER"))
+            stagger = atoi(sg);
 #define CHAIN_TPL_CASE(KV)                                                \
     case KV:                                                              \
         hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_tpl<KV>),    \
@@ -4170,7 +4182,7 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                            c->fused->d_mtaps, (long long)prod,            \
                            (long long)n_in,                               \
                            (const float2*)c->fft->d_twid, (float*)d_mag,  \
-                           (int)L);                                       \
+                           (int)L, stagger);                              \
         break;
         switch (KK) {
             CHAIN_TPL_CASE(20)
