@@ -1379,6 +1379,226 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_mfma_ubench_tpl(
                          cim[0] + cim[1] + cim[2] + cim[3]);
 }
 
+/* Incremental chain ubench: the real tile loop with components gated by
+ * MODE to isolate the integration cost (tools/mfma_ubench.py):
+ *   bit0 (1): global staging loads of the real input
+ *   bit1 (2): LDS write_half phases + their barriers
+ *   bit2 (4): deposit + in-block FFT + output writes
+ * MODE 7 == the production kernel shape; MODE 0 ~= the bare loop. */
+template <int KKD, int MODE>
+__global__ __launch_bounds__(MDFIR_BLOCK) void k_chain_ubench_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv, long long n_out, long long n_in_valid,
+    const float2* __restrict__ twid) {
+    const unsigned elemsP = MDFIR_TILE + KKD + 8;
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    const unsigned SUB = SPm / 4;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* planes = (float*)smem;
+    float* s_rtx = planes + 4u * SPm;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+    for (int i = tid; i < 4 * (KKD + 16); i += MDFIR_BLOCK) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    for (int i = tid; i < (int)(4 * SPm); i += MDFIR_BLOCK)
+        planes[i] = (float)(i & 255) * 0.001f;
+    __syncthreads();
+    const unsigned span = 3 + 4 * elemsP;
+    constexpr int NL2 =
+        (2 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float2 stgA[NL2], stgB[NL2];
+    auto load_half = [&](long long tl, int h, float2 (&stg)[NL2]) {
+        if (MODE & 1) {
+            const long long ib = tl * MDFIR_TILE * 4;
+#pragma unroll
+            for (int j = 0; j < NL2; j++) {
+                unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+                unsigned rel = 3 + 4 * (idx >> 1) + 2 * h + (idx & 1u);
+                long long g = ib + rel;
+                stg[j] = (rel < span && g < n_in_valid)
+                             ? in[g] : make_float2(0.f, 0.f);
+            }
+        } else {
+#pragma unroll
+            for (int j = 0; j < NL2; j++)
+                stg[j] = make_float2((float)(tid + j + h) * 1e-3f,
+                                     (float)(tl & 63) * 1e-3f);
+        }
+    };
+    auto write_half = [&](const float2 (&stg)[NL2]) {
+        if (MODE & 2) {
+#pragma unroll
+            for (int j = 0; j < NL2; j++) {
+                unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
+                unsigned i = idx >> 1, vloc = idx & 1u;
+                if (i < elemsP) {
+                    unsigned d = (i & 3u) * SUB + (i >> 2);
+                    planes[vloc * SPm + d] = stg[j].x;
+                    planes[(2 + vloc) * SPm + d] = stg[j].y;
+                }
+            }
+        } else { /* keep stg live without LDS traffic */
+            float acc = 0.f;
+#pragma unroll
+            for (int j = 0; j < NL2; j++) acc += stg[j].x;
+            if (acc == 1e30f) planes[tid] = acc; /* never taken */
+        }
+    };
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+    auto mfma_half = [&](int h, v4f& cre, v4f& cim) {
+#pragma unroll
+        for (int vloc = 0; vloc < 2; vloc++) {
+            const float* pre = planes + (unsigned)vloc * SPm + asub;
+            const float* pim = planes + (unsigned)(2 + vloc) * SPm + asub;
+            const int v = 2 * h + vloc;
+            float bfrag[KKD / 4];
+#pragma unroll
+            for (int s = 0; s < KKD / 4; s++)
+                bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int t = 0; t < (KKD / 4) / 4; t++) {
+                float4 ar = *(const float4*)&pre[abase + 4 * t];
+                float4 ai = *(const float4*)&pim[abase + 4 * t];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.x, bfrag[4 * t], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.x, bfrag[4 * t], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
+            }
+            __builtin_amdgcn_s_setprio(0);
+        }
+    };
+    load_half(blockIdx.x, 0, stgA);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
+        const long long out_base = tile * MDFIR_TILE;
+        v4f cre = {0.f, 0.f, 0.f, 0.f};
+        v4f cim = {0.f, 0.f, 0.f, 0.f};
+        write_half(stgA);
+        if (MODE & 2) __syncthreads();
+        load_half(tile, 1, stgB);
+        mfma_half(0, cre, cim);
+        if (MODE & 2) __syncthreads();
+        write_half(stgB);
+        if (MODE & 2) __syncthreads();
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_half(tile + gridDim.x, 0, stgA);
+        mfma_half(1, cre, cim);
+        if (MODE & 4) {
+            __syncthreads();
+            float2* ping = (float2*)planes;
+            float2* pong = ping + 1024;
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                int row = k4 * 4 + q;
+                unsigned pos = wave * 256 + 16 * row + r16;
+                ping[fft_swz(pos)] = make_float2(cre[q], cim[q]);
+            }
+            __syncthreads();
+            fft1024_block(ping, pong, twid, tid);
+            for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+                long long o = out_base + i;
+                if (o < n_out) out[o] = pong[fft_swz((unsigned)i)];
+            }
+            __syncthreads();
+        } else {
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                int row = k4 * 4 + q;
+                long long o =
+                    out_base + (long long)wave * 256 + 16 * row + r16;
+                if (o < n_out) out[o] = make_float2(cre[q], cim[q]);
+            }
+            if (MODE & 2) __syncthreads();
+        }
+    }
+}
+
+extern "C" int fsdr_chain_ubench(int mode, double* tflops, void* stream) {
+    REQUIRE_GPU();
+    const int KKD = 80;
+    const long long n_out = 1 << 24; /* decimated outputs (2^26 inputs) */
+    const long long n_in = 4 * n_out + 1024;
+    unsigned elemsP = MDFIR_TILE + KKD + 8;
+    unsigned SPm = (elemsP + 31u) & ~31u;
+    size_t lds = (4 * (size_t)SPm + 4 * (KKD + 16)) * sizeof(float);
+    float* d_taps = nullptr;
+    float2* d_in = nullptr;
+    float2* d_out = nullptr;
+    float2* d_tw = nullptr;
+    HIP_TRY(hipMalloc(&d_taps, 4 * KKD * sizeof(float)));
+    HIP_TRY(hipMemset(d_taps, 1, 4 * KKD * sizeof(float)));
+    HIP_TRY(hipMalloc(&d_in, n_in * sizeof(float2)));
+    HIP_TRY(hipMemset(d_in, 2, n_in * sizeof(float2)));
+    HIP_TRY(hipMalloc(&d_out, n_out * sizeof(float2)));
+    std::vector<float2> tw(1024);
+    for (int k = 0; k < 1024; k++) {
+        double a = -2.0 * M_PI * k / 1024.0;
+        tw[k] = make_float2((float)cos(a), (float)sin(a));
+    }
+    HIP_TRY(hipMalloc(&d_tw, 1024 * sizeof(float2)));
+    HIP_TRY(hipMemcpy(d_tw, tw.data(), 1024 * sizeof(float2),
+                      hipMemcpyHostToDevice));
+    hipStream_t st = (hipStream_t)stream;
+    int grid = 8192;
+    hipEvent_t e0, e1;
+    HIP_TRY(hipEventCreate(&e0));
+    HIP_TRY(hipEventCreate(&e1));
+#define UB_CASE(MV)                                                       \
+    case MV:                                                              \
+        hipLaunchKernelGGL(HIP_KERNEL_NAME((k_chain_ubench_tpl<80, MV>)), \
+                           dim3(grid), dim3(MDFIR_BLOCK), lds, st, d_in,  \
+                           d_out, d_taps, n_out, n_in, d_tw);             \
+        break;
+    for (int rep = 0; rep < 2; rep++) {
+        if (rep == 1) HIP_TRY(hipEventRecord(e0, st));
+        switch (mode) {
+            UB_CASE(0)
+            UB_CASE(1)
+            UB_CASE(2)
+            UB_CASE(3)
+            UB_CASE(4)
+            UB_CASE(6)
+            UB_CASE(7)
+            default:
+                set_err("mode must be in {0,1,2,3,4,6,7}");
+                return FSDR_ERR_INVALID;
+        }
+        HIP_TRY(hipGetLastError());
+    }
+    HIP_TRY(hipEventRecord(e1, st));
+    HIP_TRY(hipEventSynchronize(e1));
+    float ms = 0.f;
+    HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
+    double fl = (double)n_out * 4 * KKD * 4.0; /* 4 phases x KKD x 4 */
+    *tflops = fl / (ms * 1e-3) / 1e12;
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    (void)hipFree(d_taps);
+    (void)hipFree(d_in);
+    (void)hipFree(d_out);
+    (void)hipFree(d_tw);
+    return FSDR_OK;
+}
+
 extern "C" int fsdr_mfma_ubench(int grid, int iters, double* tflops,
                                 void* stream) {
     REQUIRE_GPU();

@@ -217,6 +217,10 @@ size_t fsdr_wlan_sync_long_run(fsdr_wlan_rx* rx, const fsdr_cf32* in,
 /* MFMA-loop microbenchmark (diagnostics; tools/mfma_ubench.py): times
  * the chain kernel's inner MFMA loop on LDS staged once. */
 int fsdr_mfma_ubench(int grid, int iters, double* tflops, void* stream);
+/* Incremental chain ubench: real tile loop with components gated by
+ * mode bits (1 staging loads, 2 LDS writes+barriers, 4 deposit+FFT);
+ * reports executed-MFMA TF/s for pipe-utilization comparison. */
+int fsdr_chain_ubench(int mode, double* tflops, void* stream);
 
 /* ---- device memory helpers (for harnesses driving the _dev paths) ---- */
 int fsdr_dev_alloc(void** d_ptr, size_t bytes);
