@@ -1029,30 +1029,64 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ap_tpl(
     }
     __syncthreads();
 
-    const unsigned span = 3 + 4 * elemsP;
-    constexpr int NL4 =
-        (4 * (MDFIR_TILE + KKD + 8) + 3 + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
-    float2 stg[NL4];
+    /* Staging by ALIGNED 4-sample groups: group q = x[4q..4q+3] = two
+     * 16B float4 loads (x is 32B-aligned at 4q when the base pointer is
+     * 16B-aligned — host guards). Sample x[4q+w] belongs to phase
+     * v=(w+1)&3 at position q (w==3) or q-1 (w<3), since
+     * P_v[i] = x[3+v+4i]. Halves the VMEM instruction count vs float2
+     * loads — the incremental ubench showed the staging loads cost ~25
+     * points of MFMA-pipe utilization (issue/latency interference). */
+    constexpr unsigned GROUPS = (MDFIR_TILE + KKD + 8) + 1;
+    constexpr int NG = (GROUPS + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float4 stqa[NG], stqb[NG];
     auto load_all = [&](long long tl) {
-        const long long ib = tl * MDFIR_TILE * 4;
+        const long long qb = tl * MDFIR_TILE;
 #pragma unroll
-        for (int j = 0; j < NL4; j++) {
-            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
-            unsigned rel = 3 + idx;
-            long long g = ib + rel;
-            stg[j] = (rel < span && g < n_in_valid)
-                         ? in[g] : make_float2(0.f, 0.f);
+        for (int j = 0; j < NG; j++) {
+            unsigned q = (unsigned)(tid + j * MDFIR_BLOCK);
+            float4 a = make_float4(0.f, 0.f, 0.f, 0.f);
+            float4 b = make_float4(0.f, 0.f, 0.f, 0.f);
+            if (q < GROUPS) {
+                long long xq = (qb + q) * 4;
+                if (xq + 3 < n_in_valid) {
+                    a = *(const float4*)&in[xq];
+                    b = *(const float4*)&in[xq + 2];
+                } else {
+                    float2 e0 = (xq < n_in_valid) ? in[xq]
+                                                  : make_float2(0.f, 0.f);
+                    float2 e1 = (xq + 1 < n_in_valid)
+                                    ? in[xq + 1] : make_float2(0.f, 0.f);
+                    float2 e2 = (xq + 2 < n_in_valid)
+                                    ? in[xq + 2] : make_float2(0.f, 0.f);
+                    float2 e3 = (xq + 3 < n_in_valid)
+                                    ? in[xq + 3] : make_float2(0.f, 0.f);
+                    a = make_float4(e0.x, e0.y, e1.x, e1.y);
+                    b = make_float4(e2.x, e2.y, e3.x, e3.y);
+                }
+            }
+            stqa[j] = a;
+            stqb[j] = b;
         }
     };
     auto write_all = [&]() {
 #pragma unroll
-        for (int j = 0; j < NL4; j++) {
-            unsigned idx = (unsigned)(tid + j * MDFIR_BLOCK);
-            unsigned v = idx & 3u, u = idx >> 2;
-            if (u < elemsP) {
-                unsigned d = v * SPm + (u & 3u) * SUB + (u >> 2);
-                planes[d] = stg[j].x;
-                planes[4u * SPm + d] = stg[j].y;
+        for (int j = 0; j < NG; j++) {
+            unsigned q = (unsigned)(tid + j * MDFIR_BLOCK);
+            if (q >= GROUPS) continue;
+            float4 a = stqa[j], b = stqb[j];
+            if (q >= 1) { /* w=0,1,2 -> phases 1,2,3 at position q-1 */
+                unsigned d1 = ((q - 1) & 3u) * SUB + ((q - 1) >> 2);
+                planes[1u * SPm + d1] = a.x;
+                planes[5u * SPm + d1] = a.y;
+                planes[2u * SPm + d1] = a.z;
+                planes[6u * SPm + d1] = a.w;
+                planes[3u * SPm + d1] = b.x;
+                planes[7u * SPm + d1] = b.y;
+            }
+            if (q < elemsP) { /* w=3 -> phase 0 at position q */
+                unsigned d0 = (q & 3u) * SUB + (q >> 2);
+                planes[0u * SPm + d0] = b.z;
+                planes[4u * SPm + d0] = b.w;
             }
         }
     };
@@ -4350,7 +4384,9 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
             return FSDR_OK;
         }
         const char* ap = getenv("FSDR_CHAIN_ALLPHASE");
-        if (ap && atoi(ap) != 0) {
+        /* the ap kernel's group staging needs a 16B-aligned input base
+         * (ring carry offsets can be 8B-odd -> fall back to halves) */
+        if (ap && atoi(ap) != 0 && ((uintptr_t)d_in & 15u) == 0) {
             unsigned SPm = (elemsP + 31u) & ~31u;
             size_t lds_ap =
                 (8 * (size_t)SPm + 4 * ((size_t)KK + 16)) * sizeof(float);
