@@ -4384,9 +4384,13 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
             return FSDR_OK;
         }
         const char* ap = getenv("FSDR_CHAIN_ALLPHASE");
-        /* the ap kernel's group staging needs a 16B-aligned input base
-         * (ring carry offsets can be 8B-odd -> fall back to halves) */
-        if (ap && atoi(ap) != 0 && ((uintptr_t)d_in & 15u) == 0) {
+        /* all-phase + aligned-group staging is the default (measured
+         * ~2% faster: fewer barriers and half the staging-load
+         * instructions; see profiles/pmc_sq_r02.txt). The ap kernel's
+         * group staging needs a 16B-aligned input base — ring carry
+         * offsets can be 8B-odd — so unaligned inputs (and
+         * FSDR_CHAIN_ALLPHASE=0) take the 2-phase-halves kernel. */
+        if ((!ap || atoi(ap) != 0) && ((uintptr_t)d_in & 15u) == 0) {
             unsigned SPm = (elemsP + 31u) & ~31u;
             size_t lds_ap =
                 (8 * (size_t)SPm + 4 * ((size_t)KK + 16)) * sizeof(float);
