@@ -2353,13 +2353,16 @@ __device__ float2* fft_pow2_fwd(float2* a, float2* b,
             float2 e0 = f2_add(x0, x2), e1 = f2_sub(x0, x2);
             float2 o0 = f2_add(x1, x3), o1 = f2_sub(x1, x3);
             float2 o1r = make_float2(o1.y, -o1.x);
+            /* one twiddle load; W^2p, W^3p derived by complex mults
+             * (unit-vector products, ~2 ulp — the 3-load version was
+             * VMEM-issue-bound inside the fused chain kernel) */
+            float2 w1 = twid[(size_t)p * tw];
+            float2 w2 = cmulf(w1, w1);
+            float2 w3 = cmulf(w2, w1);
             b[fft_swz(q + scur * (4 * p + 0))] = f2_add(e0, o0);
-            b[fft_swz(q + scur * (4 * p + 1))] =
-                cmulf(f2_add(e1, o1r), twid[(size_t)p * tw]);
-            b[fft_swz(q + scur * (4 * p + 2))] =
-                cmulf(f2_sub(e0, o0), twid[(size_t)2 * p * tw]);
-            b[fft_swz(q + scur * (4 * p + 3))] =
-                cmulf(f2_sub(e1, o1r), twid[(size_t)3 * p * tw]);
+            b[fft_swz(q + scur * (4 * p + 1))] = cmulf(f2_add(e1, o1r), w1);
+            b[fft_swz(q + scur * (4 * p + 2))] = cmulf(f2_sub(e0, o0), w2);
+            b[fft_swz(q + scur * (4 * p + 3))] = cmulf(f2_sub(e1, o1r), w3);
         }
         float2* t = a; a = b; b = t;
         scur <<= 2;
@@ -2487,21 +2490,24 @@ __global__ __launch_bounds__(256) void k_fft_stockham(
                     float2 x3 = a[fft_swz(q + scur * (p + 3 * m4))];
                     /* DIF radix-4: butterflies first, output twiddles
                      * W^p, W^2p, W^3p on frequencies 1..3 (omega4 = -i
-                     * forward, +i inverse) */
+                     * forward, +i inverse). One twiddle load; the
+                     * squares/cubes are derived (unit-vector products,
+                     * ~2 ulp). */
                     float2 e0 = f2_add(x0, x2), e1 = f2_sub(x0, x2);
                     float2 o0 = f2_add(x1, x3), o1 = f2_sub(x1, x3);
                     float2 o1r = inverse ? make_float2(-o1.y, o1.x)
                                          : make_float2(o1.y, -o1.x);
+                    float2 w1 = twid[(size_t)p * tw];
+                    if (inverse) w1.y = -w1.y;
+                    float2 w2 = cmulf(w1, w1);
+                    float2 w3 = cmulf(w2, w1);
                     b[fft_swz(q + scur * (4 * p + 0))] = f2_add(e0, o0);
                     b[fft_swz(q + scur * (4 * p + 1))] =
-                        cmul_tw(f2_add(e1, o1r), twid[(size_t)p * tw],
-                                inverse);
+                        cmulf(f2_add(e1, o1r), w1);
                     b[fft_swz(q + scur * (4 * p + 2))] =
-                        cmul_tw(f2_sub(e0, o0), twid[(size_t)2 * p * tw],
-                                inverse);
+                        cmulf(f2_sub(e0, o0), w2);
                     b[fft_swz(q + scur * (4 * p + 3))] =
-                        cmul_tw(f2_sub(e1, o1r), twid[(size_t)3 * p * tw],
-                                inverse);
+                        cmulf(f2_sub(e1, o1r), w3);
                 }
             }
             float2* t = a; a = b; b = t;
