@@ -155,19 +155,35 @@ def measure_streaming(fa, lib, chain, taps1, taps2, decim, fft_len,
     d_mag = ctypes.c_void_p()
     if lib.fsdr_dev_alloc(ctypes.byref(d_mag), out_cap * 4) != 0:
         raise RuntimeError(lib.fsdr_last_error().decode())
+    import concurrent.futures as cf
+
     import torch
     st = torch.cuda.current_stream()
     hp = ctypes.c_void_p()
     items = ctypes.c_size_t()
     dp = ctypes.c_void_p()
     got = ctypes.c_size_t()
+    pool = cf.ThreadPoolExecutor(max_workers=8)
+
+    def fill(base, nbytes):
+        # the source's buffer write, parallel over host cores (memset
+        # releases the GIL) so the leg measures the PCIe/ring bound, not
+        # one core's memset
+        nt = 8
+        step = (nbytes + nt - 1) // nt
+        futs = [pool.submit(ctypes.memset,
+                            ctypes.c_void_p(base + i * step), 0,
+                            min(step, nbytes - i * step))
+                for i in range(nt) if i * step < nbytes]
+        for f in futs:
+            f.result()
 
     def push(n):
         total = 0
         for _ in range(n):
             lib.fsdr_ring_writer_acquire(ring, ctypes.byref(hp),
                                          ctypes.byref(items))
-            ctypes.memset(hp, 0, chunk * 8)  # the source's buffer write
+            fill(hp.value, chunk * 8)
             lib.fsdr_ring_writer_commit(ring, chunk)
             lib.fsdr_ring_reader_acquire(ring, ctypes.byref(dp),
                                          ctypes.byref(got))
@@ -197,6 +213,7 @@ def measure_streaming(fa, lib, chain, taps1, taps2, decim, fft_len,
     finally:
         lib.fsdr_dev_free(d_mag)
         lib.fsdr_ring_destroy(ring)
+        pool.shutdown(wait=False)
 
 
 def measure_config3(fa, lib, torch):
