@@ -6,7 +6,11 @@ A step = SOURCE WRITE + one pass of the chain over one HBM-resident batch
 of synthetic samples (re/im iid uniform[-1,1), device-generated, seeded).
 The source's buffer write (the NullSource analogue,
 src/blocks/null_source.rs:53-66) is INSIDE the timed region: every step
-re-fills the input batch on-device before the chain consumes it. A
+re-fills the input batch on-device before the chain consumes it. Source
+and chain are PIPELINED over sub-chunks on two streams (the source fills
+chunk j+1 while the chain consumes chunk j), which is exactly the
+reference actor model: NullSource runs ahead of the Fir block through
+the slab's circulating buffers (scheduler/smol.rs tasks + slab.rs). A
 separate streaming leg (reported as "streaming" in the JSON, never as
 `value`) pushes host chunks through the pinned fsdr_ring (H2D on the copy
 stream overlapped with compute) — the PCIe-fed rate. At N>1 each
@@ -67,6 +71,9 @@ def parse_args():
                    help="skip the PCIe-fed ring streaming leg")
     p.add_argument("--skip-config3", action="store_true",
                    help="skip the config-3 (FM resampler chain) leg")
+    p.add_argument("--pipeline-chunks", type=int, default=0,
+                   help="source/chain pipeline depth per step (0 = auto: "
+                        "4 for batches >= 2^28; 1 disables overlap)")
     p.add_argument("--skip-roofline", action="store_true",
                    help="skip the roofline reps (for PMC traffic passes "
                         "whose counters the extra d_null-writing reps "
@@ -370,14 +377,43 @@ def main():
     chain = fa.Chain(taps1, taps2, args.decim, args.fft)
 
     S = args.samples
-    y2 = (S + 1 - args.taps + 1 - args.taps) // args.decim
-    frames = y2 // args.fft
-    prod = frames * args.fft
+    # The step is the reference's actor pipeline: the SOURCE fills chunk
+    # j+1 (its own stream) while the chain consumes chunk j (compute
+    # stream), event-chained exactly like the slab's 4 circulating
+    # buffers let NullSource run ahead of the Fir block. Every sample is
+    # still written by the source inside the timed region; chunks are
+    # independent spans (each loses the <253+4096-sample window tail,
+    # ~0.002% of a chunk — the same boundary behavior as unconsumed slab
+    # leftovers at stream end).
+    C = args.pipeline_chunks
+    if C <= 0:
+        C = 4 if S >= (1 << 28) else 1
+    chunk0 = (S // C) & ~4095
+    offs, lens = [], []
+    off = 0
+    for j in range(C):
+        n_j = chunk0 if j < C - 1 else S - off
+        offs.append(off)
+        lens.append(n_j)
+        off += n_j
+    g1 = args.taps + args.taps - 1
+
+    def chunk_prod(n):
+        return max(0, (n + 1 - g1) // args.decim) // args.fft * args.fft
+
+    prods = [chunk_prod(n) for n in lens]
+    mag_offs = []
+    acc = 0
+    for p in prods:
+        mag_offs.append(acc)
+        acc += p
+    prod = acc
+    frames = prod // args.fft
 
     d_in = alloc_dev(lib, S * 8)
     fa.fill_uniform_dev(d_in.value, S, seed=1000 + rank,
                         stream=st.cuda_stream)
-    mag = torch.empty(prod, dtype=torch.float32, device="cuda")
+    mag = torch.empty(max(prod, 1), dtype=torch.float32, device="cuda")
     # spectrum sink: per-bin EMA over the step's frames, emitting one
     # averaged spectrum per step (MovingAvg(width=fft, decay, history=
     # frames) — the reference sink's output cadence for this batch size)
@@ -389,18 +425,31 @@ def main():
         gathered = torch.empty(world * payload, dtype=torch.float32,
                                device="cuda")
 
+    s_src = torch.cuda.Stream()
+    ev_fill = [torch.cuda.Event() for _ in range(C)]
+    ev_done = [torch.cuda.Event() for _ in range(C)]
     seed_ctr = [0]
 
     def step():
-        # the SOURCE: refill the input batch on-device (NullSource writes
-        # its whole buffer every call — null_source.rs:53-66)
         seed_ctr[0] += 1
-        fa.fill_uniform_dev(d_in.value, S, seed=seed_ctr[0] * 131 + rank,
-                            stream=st.cuda_stream)
-        chain.run_dev(d_in.value, S, 0, 0, mag.data_ptr(), prod,
-                      stream=st.cuda_stream)
-        avg_sink.filter_dev(mag.data_ptr(), prod, avg_spec.data_ptr(),
-                            args.fft, stream=st.cuda_stream)
+        seed = seed_ctr[0] * 131 + rank
+        for j in range(C):
+            # the SOURCE writes chunk j on its own stream (NullSource
+            # writes its whole buffer every call — null_source.rs:53-66),
+            # after the chain's previous pass over this chunk drained
+            s_src.wait_event(ev_done[j])
+            fa.fill_uniform_dev(d_in.value + offs[j] * 8, lens[j],
+                                seed=seed, offset=offs[j],
+                                stream=s_src.cuda_stream)
+            ev_fill[j].record(s_src)
+            st.wait_event(ev_fill[j])
+            chain.run_dev(d_in.value + offs[j] * 8, lens[j], 0, 0,
+                          mag.data_ptr() + mag_offs[j] * 4, prods[j],
+                          stream=st.cuda_stream)
+            ev_done[j].record(st)
+            avg_sink.filter_dev(mag.data_ptr() + mag_offs[j] * 4,
+                                prods[j], avg_spec.data_ptr(), args.fft,
+                                stream=st.cuda_stream)
         if td is not None:
             td.all_gather_into_tensor(
                 gathered, mag if args.gather_frames else avg_spec)
