@@ -7,10 +7,11 @@ of synthetic samples (re/im iid uniform[-1,1), device-generated, seeded).
 The source's buffer write (the NullSource analogue,
 src/blocks/null_source.rs:53-66) is INSIDE the timed region: every step
 re-fills the input batch on-device before the chain consumes it. Source
-and chain are PIPELINED over sub-chunks on two streams (the source fills
-chunk j+1 while the chain consumes chunk j), which is exactly the
-reference actor model: NullSource runs ahead of the Fir block through
-the slab's circulating buffers (scheduler/smol.rs tasks + slab.rs). A
+and chain CAN be pipelined over sub-chunks on two streams
+(--pipeline-chunks N, the reference actor model: NullSource running
+ahead of the Fir block through the slab's circulating buffers) — but on
+this workload both stages share HBM and overlap starves the chain, so
+the default is serial (see the step comment). A
 separate streaming leg (reported as "streaming" in the JSON, never as
 `value`) pushes host chunks through the pinned fsdr_ring (H2D on the copy
 stream overlapped with compute) — the PCIe-fed rate. At N>1 each
@@ -72,8 +73,9 @@ def parse_args():
     p.add_argument("--skip-config3", action="store_true",
                    help="skip the config-3 (FM resampler chain) leg")
     p.add_argument("--pipeline-chunks", type=int, default=0,
-                   help="source/chain pipeline depth per step (0 = auto: "
-                        "4 for batches >= 2^28; 1 disables overlap)")
+                   help="source/chain pipeline depth per step (0 = auto "
+                        "= 1: overlap measured slower — fill and chain "
+                        "contend for HBM; see step comment)")
     p.add_argument("--skip-roofline", action="store_true",
                    help="skip the roofline reps (for PMC traffic passes "
                         "whose counters the extra d_null-writing reps "
@@ -385,9 +387,15 @@ def main():
     # independent spans (each loses the <253+4096-sample window tail,
     # ~0.002% of a chunk — the same boundary behavior as unconsumed slab
     # leftovers at stream end).
+    # Measured: overlap does NOT pay here — the fill is write-bandwidth
+    # bound (~5.7 TB/s) and the chain read-needs ~2.9 TB/s, so running
+    # them concurrently over the SHARED HBM starves the chain below its
+    # compute rate (pc=4 was ~5% slower than serial). Auto = 1; the
+    # pipeline stays available for configs where the consumer is
+    # compute-bound with bandwidth slack.
     C = args.pipeline_chunks
     if C <= 0:
-        C = 4 if S >= (1 << 28) else 1
+        C = 1
     chunk0 = (S // C) & ~4095
     offs, lens = [], []
     off = 0
