@@ -143,7 +143,9 @@ def measure_chain_roofline(fa, torch, chain, d_in, n_samples, taps1,
         "unit": "TFLOP/s",
         "frac": round(achieved_tf / FP32_PEAK_TFLOPS, 4),
         "traffic": traffic,
-        "kernel": "k_decim4_fft_mfma_tpl<80> (fused fir+decim+fft+mag)",
+        "kernel": "k_decim4_fft_mfma_ap_tpl<80> (fused fir+decim+fft+mag; "
+                  "all-phase aligned-group staging — halves variant for "
+                  "unaligned ring carries)",
         "ms_per_launch": round(ms, 4),
     }
 

@@ -283,6 +283,41 @@ def test_fft_bluestein_inverse_shift_normalize(gpu, oracle_lib):
         assert rel < 2e-4, (kw, rel)
 
 
+def test_chain_allphase_vs_halves_kernels(gpu, oracle_lib):
+    """The two fused-kernel variants (all-phase aligned-group staging =
+    default for 16B-aligned inputs; 2-phase-halves = the unaligned ring
+    fallback) compute the same MFMA/K order — outputs must agree to
+    float equality on identical input."""
+    r = rng(4400)
+    t1 = r.uniform(-1, 1, 127).astype(np.float32)
+    t2 = r.uniform(-1, 1, 127).astype(np.float32)
+    n_in = 300000
+    x = cplx(r, n_in)
+    lib = gpu.lib()
+    d_in = ctypes.c_void_p()
+    d_out = ctypes.c_void_p()
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_in), n_in * 8) == 0
+    assert lib.fsdr_dev_alloc(ctypes.byref(d_out), n_in * 8) == 0
+    try:
+        lib.fsdr_memcpy_h2d(d_in, ctypes.c_void_p(x.ctypes.data), n_in * 8)
+        outs = {}
+        for env in ("1", "0"):
+            os.environ["FSDR_CHAIN_ALLPHASE"] = env
+            ch = gpu.Chain(t1, t2, 4, 1024)
+            cons, prod = ch.run_dev(d_in.value, n_in, d_out.value, n_in)
+            gpu.synchronize()
+            h = np.zeros(prod, np.complex64)
+            lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
+                                prod * 8)
+            outs[env] = h
+        os.environ.pop("FSDR_CHAIN_ALLPHASE", None)
+        assert outs["1"].size == outs["0"].size
+        assert_close(outs["1"], outs["0"], 1e-6)
+    finally:
+        lib.fsdr_dev_free(d_in)
+        lib.fsdr_dev_free(d_out)
+
+
 def test_chain_non_pow2_fft_vs_oracle(gpu, oracle_lib):
     """Chain with a non-pow2 FFT length: the split path runs the
     Bluestein FFT; outputs equal the two-stage oracle (which falls back
