@@ -2576,24 +2576,40 @@ __device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
     return x ^ (x >> 31);
 }
 
+/* lowbias32 (Wellons): full-avalanche 32-bit hash — two 32-bit mults
+ * per 32 output bits vs splitmix64's two 64-bit mults per 64 (64-bit
+ * integer mults are multi-op on CDNA4; the fill was ALU-capped below
+ * write bandwidth). */
+__device__ __forceinline__ uint32_t lowbias32(uint32_t x) {
+    x ^= x >> 16;
+    x *= 0x7FEB352Du;
+    x ^= x >> 15;
+    x *= 0x846CA68Bu;
+    x ^= x >> 16;
+    return x;
+}
+
 __global__ void k_fill_uniform_cf32(float2* __restrict__ out, long long n,
                                     uint64_t seed, uint64_t offset) {
     /* pair of samples per iteration, one float4 store (the source write
-     * is inside the bench's timed region — keep it at write bandwidth) */
+     * is inside the bench's timed region — keep it at write bandwidth).
+     * Counter-based: deterministic by (seed, element index). */
+    const uint32_t sm =
+        (uint32_t)seed * 0x9E3779B9u ^ (uint32_t)(seed >> 32);
     long long np = n >> 1;
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long p = blockIdx.x * (long long)blockDim.x + threadIdx.x;
          p < np; p += stride) {
         long long i = 2 * p;
-        uint64_t h0 =
-            splitmix64(seed ^ (0x5D5D5D5Dull + offset + (uint64_t)i));
-        uint64_t h1 =
-            splitmix64(seed ^ (0x5D5D5D5Dull + offset + (uint64_t)i + 1));
+        uint64_t c64 = 0x5D5D5D5Dull + offset + (uint64_t)i;
+        uint32_t c = (uint32_t)c64 ^ (uint32_t)(c64 >> 32) * 0x85EBCA6Bu;
+        uint32_t r0 = lowbias32(c * 4u + sm);
+        uint32_t r1 = lowbias32(c * 4u + 1u + sm);
+        uint32_t r2 = lowbias32(c * 4u + 2u + sm);
+        uint32_t r3 = lowbias32(c * 4u + 3u + sm);
         const float s = 2.0f / 16777216.0f;
-        float4 v = make_float4(((uint32_t)h0 >> 8) * s - 1.0f,
-                               ((uint32_t)(h0 >> 32) >> 8) * s - 1.0f,
-                               ((uint32_t)h1 >> 8) * s - 1.0f,
-                               ((uint32_t)(h1 >> 32) >> 8) * s - 1.0f);
+        float4 v = make_float4((r0 >> 8) * s - 1.0f, (r1 >> 8) * s - 1.0f,
+                               (r2 >> 8) * s - 1.0f, (r3 >> 8) * s - 1.0f);
         *(float4*)&out[i] = v;
     }
     if ((n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
