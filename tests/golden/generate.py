@@ -19,7 +19,8 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 def main():
     rng = np.random.default_rng(0x5D5D5D5D)
     out = {}
-    for n in (64, 256, 1024):
+    # pow2 (Stockham kernel) and non-pow2 (Bluestein path) lengths
+    for n in (64, 256, 1024, 60, 100, 1000):
         x = (rng.uniform(-1, 1, (4, n)) +
              1j * rng.uniform(-1, 1, (4, n))).astype(np.complex64)
         X = np.fft.fft(x.astype(np.complex128), axis=1)

@@ -16,7 +16,7 @@ def golden():
     return np.load(GOLDEN)
 
 
-@pytest.mark.parametrize("n", [64, 256, 1024])
+@pytest.mark.parametrize("n", [64, 256, 1024, 60, 100, 1000])
 def test_oracle_fft_vs_golden(oracle_lib, golden, n):
     o = oracle_lib
     xs = golden[f"fft{n}_in"]

@@ -198,7 +198,8 @@ def test_fft_parity_vs_oracle(gpu, oracle_lib, n):
 
 def test_fft_vs_golden(gpu):
     g = np.load(GOLDEN)
-    for n in (64, 256, 1024):
+    # pow2 -> Stockham kernel; non-pow2 -> Bluestein path
+    for n in (64, 256, 1024, 60, 100, 1000):
         xs, refs = g[f"fft{n}_in"], g[f"fft{n}_out"]
         got, c, p, s = gpu.Fft(n).filter(xs.ravel(), xs.size)
         got = got.reshape(xs.shape)
