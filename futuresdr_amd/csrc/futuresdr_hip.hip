@@ -468,15 +468,10 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
     long long n_out, long long n_in_valid) {
     static_assert(KK % 4 == 0, "KK must be a multiple of 4");
     const unsigned elems = MFIR_TILE + KK + 8;
-    const unsigned SP = (elems + 31u) & ~31u;
-    /* sub-plane split by element%4: a lane's K-walk (ab+4s) becomes
-     * stride-1, so one ds_read_b128 feeds 4 MFMA K-steps (same trick as
-     * the decim/chain kernels; see profiles/pmc_sq_r02.txt). */
-    const unsigned SUB = SP / 4;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     float* s_re = (float*)smem;
-    float* s_im = s_re + SP;
-    float* s_rtx = s_im + SP; /* 15 zeros + rt[KK] + 1 */
+    float* s_im = s_re + ((elems + 31u) & ~31u);
+    float* s_rtx = s_im + ((elems + 31u) & ~31u); /* 15 zeros + rt[KK] + 1 */
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -523,51 +518,25 @@ __global__ __launch_bounds__(MFIR_BLOCK) void k_fir_mfma_tpl(
         for (int j = 0; j < NL; j++) {
             unsigned i = 2 * (tid + j * MFIR_BLOCK);
             if (i < elems) {
-                unsigned d0 = (i & 3u) * SUB + (i >> 2);
-                unsigned d1 = ((i + 1) & 3u) * SUB + ((i + 1) >> 2);
-                s_re[d0] = stg[j].x;
-                s_re[d1] = stg[j].z;
-                s_im[d0] = stg[j].y;
-                s_im[d1] = stg[j].w;
+                unsigned d = mfma_swz(i);
+                *(float2*)&s_re[d] = make_float2(stg[j].x, stg[j].z);
+                *(float2*)&s_im[d] = make_float2(stg[j].y, stg[j].w);
             }
         }
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MFIR_TILE < n_out)
             load_tile(tile + gridDim.x);
 
-        const unsigned abase = (unsigned)wave * 64 + 4u * r16;
-        const float* pre = s_re + (unsigned)k4 * SUB;
-        const float* pim = s_im + (unsigned)k4 * SUB;
+        const unsigned ab = (unsigned)wave * 256 + 16u * r16 + k4;
         /* one accumulator pair, re/im alternating: same-C spacing = 2
          * MFMA issues = 64 cyc/SIMD >= the 40-cyc dependent latency
          * (extra pairs measured slower — register pressure only) */
         v4f cre = {0.f, 0.f, 0.f, 0.f};
         v4f cim = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int t = 0; t < (KK / 4) / 4; t++) {
-            float4 ar = *(const float4*)&pre[abase + 4 * t];
-            float4 ai = *(const float4*)&pim[abase + 4 * t];
-            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(ar.x, bfrag[4 * t],
-                                                       cre, 0, 0, 0);
-            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(ai.x, bfrag[4 * t],
-                                                       cim, 0, 0, 0);
-            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
-            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
-            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
-            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
-            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
-            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
-        }
-#pragma unroll
-        for (int s = (KK / 4) & ~3; s < KK / 4; s++) {
-            float a_re = pre[abase + s];
-            float a_im = pim[abase + s];
+        for (int s = 0; s < KK / 4; s++) {
+            float a_re = s_re[mfma_swz(ab + 4 * s)];
+            float a_im = s_im[mfma_swz(ab + 4 * s)];
             cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
                                                        0, 0, 0);
             cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
