@@ -730,13 +730,14 @@ def test_pfb_channelizer_oversample_parity(gpu, oracle_lib, oversample):
     assert_close(got, ref, 1e-5)
 
 
-def test_pfb_channelizer_streaming(gpu, oracle_lib):
+@pytest.mark.parametrize("over", [1.0, 2.0, 4.0])
+def test_pfb_channelizer_streaming(gpu, oracle_lib, over):
     """Arbitrary chunk sizes through the stateful streaming path; the
     caller re-presents the unconsumed tail (slab semantics). The
     concatenated channel outputs equal the one-shot oracle."""
-    N, over = 8, 2.0
+    N = 8
     D = int(N / over)
-    r = rng(167)
+    r = rng(167 + D)
     taps = r.uniform(-1, 1, 64).astype(np.float32)
     n_total = 6000
     x = cplx(r, n_total)
