@@ -136,6 +136,34 @@ def test_fg_spectrum_sink_with_moving_avg(gpu, oracle_lib):
     assert_close(got, ref, 1e-4)
 
 
+def test_fg_xlating_fir_in_graph(gpu, oracle_lib):
+    """VectorSource -> XlatingFir -> VectorSink through the native actor
+    loop: the block's stateful rotator phase must advance correctly
+    across the driver's chunked work() calls, matching the single-shot
+    oracle composition (xlating_fir.rs:27-94 semantics)."""
+    r = rng(191)
+    decim, offset, fs = 4, 12_000.0, 1_000_000.0
+    taps = r.uniform(-1, 1, 63).astype(np.float32)
+    n = 120_000
+    x = cplx(r, n)
+    import futuresdr_amd as fa
+    fg = fa.Flowgraph()
+    src = fg.vector_source(x)
+    xl = fg.filter(fa.XlatingFir(taps, decim, offset, fs))
+    snk = fg.vector_sink()
+    fg.connect(src, xl, snk)
+    fg.run()
+    got = fg.sink_data(snk)
+    i = np.arange(taps.size, dtype=np.float32)
+    ang = i * np.float32(2 * np.pi) * np.float32(offset) / np.float32(fs)
+    bpf = (np.cos(ang) + 1j * np.sin(ang)).astype(np.complex64) * taps
+    ref, co, po, so = oracle_lib.decim_fir_ccf32(decim, bpf, x, n)
+    theta = -2 * np.pi * offset * decim / fs
+    ref_rot, _ = oracle_lib.rotator(theta, ref)
+    assert got.size == ref_rot.size
+    assert_close(got, ref_rot, 1e-3)  # oracle rotator drift over 30k
+
+
 def test_fg_config3_resampler_chain(gpu, oracle_lib):
     """BASELINE configs[2] shape with the polyphase resampler:
     VectorSource -> Fir -> Resampler(1,4) -> Fft(256) -> VectorSink."""
