@@ -158,12 +158,17 @@ def test_fg_xlating_fir_in_graph(gpu, oracle_lib):
     ang = i * np.float32(2 * np.pi) * np.float32(offset) / np.float32(fs)
     bpf = (np.cos(ang) + 1j * np.sin(ang)).astype(np.complex64) * taps
     ref, co, po, so = oracle_lib.decim_fir_ccf32(decim, bpf, x, n)
-    theta = -2 * np.pi * offset * decim / fs
-    # ideal closed-form rotation (the GPU computes closed-form phases;
-    # the oracle's ITERATED rotator drifts O(n*eps) ~ 4e-3 over 30k
-    # outputs, so the ideal — not the drifting oracle — is the pin here)
+    # the block stores the f32-rounded phase increment (like the
+    # reference's f32 Rotator) and rotates ideally BY THAT increment;
+    # pin against the f32-theta ideal — an f64-theta ideal diverges by
+    # k*ulp(theta) ~ 5e-4 rad by output 30k, and the oracle's iterated
+    # rotator drifts even more over this span.
+    # replicate the create's sequential f32 arithmetic exactly
+    theta32 = np.float32(np.float32(np.float32(
+        np.float32(-6.2831853071795864769) * np.float32(offset))
+        * np.float32(decim)) / np.float32(fs))
     ref_rot = ref * np.exp(
-        1j * theta * (np.arange(ref.size, dtype=np.float64) + 1))
+        1j * float(theta32) * (np.arange(ref.size, dtype=np.float64) + 1))
     assert got.size == ref_rot.size
     assert_close(got, ref_rot.astype(np.complex64), 5e-4)
 
