@@ -101,7 +101,8 @@ def alloc_dev(lib, bytes_):
 
 
 def measure_chain_roofline(fa, torch, chain, d_in, n_samples, taps1,
-                           taps2, decim, fft_len, traffic_file):
+                           taps2, decim, fft_len, traffic_file,
+                           d_mag=0, mag_cap=0):
     """Dominant-kernel roofline. The default chain is ONE kernel
     (k_decim4_fft_mfma_tpl<80>: the algebraically-fused 253-tap
     decimating filter with an in-block 1024-pt FFT — DESIGN.md §d), so
@@ -115,7 +116,7 @@ def measure_chain_roofline(fa, torch, chain, d_in, n_samples, taps1,
     frames = produced // fft_len
     prod = frames * fft_len
     for _ in range(3):
-        chain.run_dev(d_in.value, n_samples, 0, 0, 0, 0,
+        chain.run_dev(d_in.value, n_samples, 0, 0, d_mag, mag_cap,
                       stream=st.cuda_stream)
     torch.cuda.synchronize()
     reps = 20
@@ -123,7 +124,8 @@ def measure_chain_roofline(fa, torch, chain, d_in, n_samples, taps1,
     ev1 = torch.cuda.Event(enable_timing=True)
     ev0.record(st)
     for _ in range(reps):
-        chain.run_dev(d_in.value, n_samples, 0, 0, 0, 0,
+        # same mag-only shape as the timed step (spectra skipped)
+        chain.run_dev(d_in.value, n_samples, 0, 0, d_mag, mag_cap,
                       stream=st.cuda_stream)
     ev1.record(st)
     torch.cuda.synchronize()
@@ -493,7 +495,8 @@ def main():
         if not args.skip_roofline:
             roofline = measure_chain_roofline(fa, torch, chain, d_in, S,
                                               taps1, taps2, args.decim,
-                                              args.fft, args.traffic_file)
+                                              args.fft, args.traffic_file,
+                                              mag.data_ptr(), prod)
         cpu_baseline = None
         if n_gpus == 1 and not args.skip_cpu_baseline:
             log("measuring CPU baseline (oracle chain, all cores)...")
