@@ -1178,6 +1178,246 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ap_tpl(
     }
 }
 
+/* Software-pipelined variant (FSDR_CHAIN_WS=1, fft_len==1024 only):
+ * tile t's FFT stages are interleaved between tile t+1's MFMA vloc
+ * groups, so the FFT's VALU/LDS work issues inside the MFMA pipe's
+ * 32-cyc/SIMD instruction shadow instead of serializing after it (the
+ * incremental ubench priced the serial FFT at ~1800 CU-cyc/tile). The
+ * FFT runs IN-PLACE (DIF radix-4, digit-reversed output read) in a
+ * dedicated 8 KB LDS strip so staging planes stay resident; ~45.6 KB
+ * LDS -> 3 blocks/CU (vs the ap kernel's 4). */
+__device__ __forceinline__ unsigned rev4_10(unsigned i) {
+    /* reverse 5 base-4 digits of a 10-bit index */
+    unsigned r = 0;
+#pragma unroll
+    for (int d = 0; d < 5; d++) {
+        r = (r << 2) | (i & 3u);
+        i >>= 2;
+    }
+    return r;
+}
+
+template <int KKD>
+__global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
+    const float2* __restrict__ in, float2* __restrict__ out,
+    const float* __restrict__ rtv, long long n_out, long long n_in_valid,
+    const float2* __restrict__ twid /* 1024-entry forward table */,
+    float* __restrict__ mag_out) {
+    static_assert(KKD % 4 == 0, "KKD must be a multiple of 4");
+    const unsigned elemsP = MDFIR_TILE + KKD + 8;
+    const unsigned SPm = (elemsP + 31u) & ~31u;
+    const unsigned SUB = SPm / 4;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    float* planes = (float*)smem;            /* [8][SPm] */
+    float* s_rtx = planes + 8u * SPm;        /* [4][KKD+16] */
+    float2* fbuf = (float2*)(s_rtx + 4 * (KKD + 16)); /* 1024 f2, in-place */
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int r16 = lane & 15;
+    const int k4 = lane >> 4;
+
+    for (int i = tid; i < 4 * (KKD + 16); i += MDFIR_BLOCK) {
+        int v = i / (KKD + 16), t = i % (KKD + 16);
+        s_rtx[i] = (t >= 15 && t < 15 + KKD) ? rtv[v * KKD + (t - 15)] : 0.f;
+    }
+    __syncthreads();
+
+    constexpr unsigned GROUPS = (MDFIR_TILE + KKD + 8) + 1;
+    constexpr int NG = (GROUPS + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
+    float4 stqa[NG], stqb[NG];
+    auto load_all = [&](long long tl) {
+        const long long qb = tl * MDFIR_TILE;
+#pragma unroll
+        for (int j = 0; j < NG; j++) {
+            unsigned q = (unsigned)(tid + j * MDFIR_BLOCK);
+            float4 a = make_float4(0.f, 0.f, 0.f, 0.f);
+            float4 b = make_float4(0.f, 0.f, 0.f, 0.f);
+            if (q < GROUPS) {
+                long long xq = (qb + q) * 4;
+                if (xq + 3 < n_in_valid) {
+                    a = *(const float4*)&in[xq];
+                    b = *(const float4*)&in[xq + 2];
+                } else {
+                    float2 e0 = (xq < n_in_valid) ? in[xq]
+                                                  : make_float2(0.f, 0.f);
+                    float2 e1 = (xq + 1 < n_in_valid)
+                                    ? in[xq + 1] : make_float2(0.f, 0.f);
+                    float2 e2 = (xq + 2 < n_in_valid)
+                                    ? in[xq + 2] : make_float2(0.f, 0.f);
+                    float2 e3 = (xq + 3 < n_in_valid)
+                                    ? in[xq + 3] : make_float2(0.f, 0.f);
+                    a = make_float4(e0.x, e0.y, e1.x, e1.y);
+                    b = make_float4(e2.x, e2.y, e3.x, e3.y);
+                }
+            }
+            stqa[j] = a;
+            stqb[j] = b;
+        }
+    };
+    auto write_all = [&]() {
+#pragma unroll
+        for (int j = 0; j < NG; j++) {
+            unsigned q = (unsigned)(tid + j * MDFIR_BLOCK);
+            if (q >= GROUPS) continue;
+            float4 a = stqa[j], b = stqb[j];
+            if (q >= 1) {
+                unsigned d1 = ((q - 1) & 3u) * SUB + ((q - 1) >> 2);
+                planes[1u * SPm + d1] = a.x;
+                planes[5u * SPm + d1] = a.y;
+                planes[2u * SPm + d1] = a.z;
+                planes[6u * SPm + d1] = a.w;
+                planes[3u * SPm + d1] = b.x;
+                planes[7u * SPm + d1] = b.y;
+            }
+            if (q < elemsP) {
+                unsigned d0 = (q & 3u) * SUB + (q >> 2);
+                planes[0u * SPm + d0] = b.z;
+                planes[4u * SPm + d0] = b.w;
+            }
+        }
+    };
+    const unsigned abase = (unsigned)wave * 64 + 4u * r16;
+    const unsigned asub = (unsigned)k4 * SUB;
+    auto mfma_vloc = [&](int v, v4f& cre, v4f& cim) {
+        const float* pre = planes + (unsigned)v * SPm + asub;
+        const float* pim = planes + (4u + v) * SPm + asub;
+        float bfrag[KKD / 4];
+#pragma unroll
+        for (int s = 0; s < KKD / 4; s++)
+            bfrag[s] = s_rtx[v * (KKD + 16) + 15 + 4 * s + k4 - r16];
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int t = 0; t < (KKD / 4) / 4; t++) {
+            float4 ar = *(const float4*)&pre[abase + 4 * t];
+            float4 ai = *(const float4*)&pim[abase + 4 * t];
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(ar.x, bfrag[4 * t],
+                                                       cre, 0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(ai.x, bfrag[4 * t],
+                                                       cim, 0, 0, 0);
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                ar.y, bfrag[4 * t + 1], cre, 0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                ai.y, bfrag[4 * t + 1], cim, 0, 0, 0);
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                ar.z, bfrag[4 * t + 2], cre, 0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                ai.z, bfrag[4 * t + 2], cim, 0, 0, 0);
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                ar.w, bfrag[4 * t + 3], cre, 0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
+        }
+#pragma unroll
+        for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
+            float a_re = pre[abase + s];
+            float a_im = pim[abase + s];
+            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
+                                                       0, 0, 0);
+            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
+                                                       0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+    };
+    /* one in-place DIF radix-4 stage: 256 butterflies, 1 per thread.
+     * Stage s: sub-length L = 1024>>(2s), quarter m = L/4; y_r stored
+     * back at q + r*m with twiddles W_L^{rq} (derived from one load). */
+    auto fft_stage = [&](int s) {
+        const unsigned L = 1024u >> (2 * s);
+        const unsigned m = L >> 2;
+        unsigned bf = (unsigned)tid;
+        unsigned blk = bf / m, q = bf - blk * m;
+        unsigned base = blk * L + q;
+        float2 x0 = fbuf[fft_swz(base)];
+        float2 x1 = fbuf[fft_swz(base + m)];
+        float2 x2 = fbuf[fft_swz(base + 2 * m)];
+        float2 x3 = fbuf[fft_swz(base + 3 * m)];
+        float2 t0 = f2_add(x0, x2), t1 = f2_sub(x0, x2);
+        float2 t2 = f2_add(x1, x3), t3 = f2_sub(x1, x3);
+        float2 t3r = make_float2(t3.y, -t3.x); /* -i * t3 (forward) */
+        float2 w1 = twid[(size_t)q << (2 * s)];
+        float2 w2 = cmulf(w1, w1);
+        float2 w3 = cmulf(w2, w1);
+        fbuf[fft_swz(base)] = f2_add(t0, t2);
+        fbuf[fft_swz(base + m)] = cmulf(f2_add(t1, t3r), w1);
+        fbuf[fft_swz(base + 2 * m)] = cmulf(f2_sub(t0, t2), w2);
+        fbuf[fft_swz(base + 3 * m)] = cmulf(f2_sub(t1, t3r), w3);
+    };
+
+    v4f cre = {0.f, 0.f, 0.f, 0.f};
+    v4f cim = {0.f, 0.f, 0.f, 0.f};
+    bool have_prev = false;
+    long long prev_base = 0;
+    load_all(blockIdx.x);
+    for (long long tile = blockIdx.x;
+         tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
+        write_all();
+        if (have_prev) { /* deposit C(prev) into the FFT strip */
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                int row = k4 * 4 + q;
+                unsigned pos = wave * 256 + 16 * row + r16;
+                fbuf[fft_swz(pos)] = make_float2(cre[q], cim[q]);
+            }
+        }
+        __syncthreads();
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_all(tile + gridDim.x);
+        cre = (v4f){0.f, 0.f, 0.f, 0.f};
+        cim = (v4f){0.f, 0.f, 0.f, 0.f};
+        mfma_vloc(0, cre, cim);
+        if (have_prev) fft_stage(0);
+        __syncthreads();
+        mfma_vloc(1, cre, cim);
+        if (have_prev) fft_stage(1);
+        __syncthreads();
+        mfma_vloc(2, cre, cim);
+        if (have_prev) fft_stage(2);
+        __syncthreads();
+        mfma_vloc(3, cre, cim);
+        if (have_prev) fft_stage(3);
+        __syncthreads();
+        if (have_prev) {
+            fft_stage(4);
+            __syncthreads();
+            for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+                long long o = prev_base + i;
+                if (o < n_out) {
+                    float2 v = fbuf[fft_swz(rev4_10((unsigned)i))];
+                    if (out) out[o] = v;
+                    if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+                }
+            }
+        }
+        prev_base = tile * MDFIR_TILE;
+        have_prev = true;
+        __syncthreads();
+    }
+    /* epilogue: FFT + output of the final tile */
+    if (have_prev) {
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            unsigned pos = wave * 256 + 16 * row + r16;
+            fbuf[fft_swz(pos)] = make_float2(cre[q], cim[q]);
+        }
+        __syncthreads();
+        for (int s = 0; s < 5; s++) {
+            fft_stage(s);
+            __syncthreads();
+        }
+        for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+            long long o = prev_base + i;
+            if (o < n_out) {
+                float2 v = fbuf[fft_swz(rev4_10((unsigned)i))];
+                if (out) out[o] = v;
+                if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+            }
+        }
+    }
+}
+
 template <int KKD>
 __global__ __launch_bounds__(512) void k_decim4_fft_mfma2_tpl(
     const float2* __restrict__ in, float2* __restrict__ out,
@@ -4402,6 +4642,37 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
                                (long long)n_in,
                                (const float2*)c->fft->d_twid,
                                (float*)d_mag);
+            HIP_TRY(hipGetLastError());
+            return FSDR_OK;
+        }
+        const char* ws = getenv("FSDR_CHAIN_WS");
+        if (ws && atoi(ws) != 0 && L == 1024 &&
+            ((uintptr_t)d_in & 15u) == 0) {
+            /* experiment: software-pipelined FFT-in-MFMA-shadow variant */
+            unsigned SPm = (elemsP + 31u) & ~31u;
+            size_t lds_ws = (8 * (size_t)SPm + 4 * ((size_t)KK + 16)) *
+                                sizeof(float) +
+                            1024 * sizeof(float2);
+#define CHAIN_WS_CASE(KV)                                                 \
+    case KV:                                                              \
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(k_decim4_fft_mfma_ws_tpl<KV>), \
+                           dim3(grid), dim3(MDFIR_BLOCK), lds_ws, st,     \
+                           (const float2*)d_in, spec_dst,                 \
+                           c->fused->d_mtaps, (long long)prod,            \
+                           (long long)n_in,                               \
+                           (const float2*)c->fft->d_twid, (float*)d_mag); \
+        break;
+            switch (KK) {
+                CHAIN_WS_CASE(20)
+                CHAIN_WS_CASE(32)
+                CHAIN_WS_CASE(48)
+                CHAIN_WS_CASE(80)
+                CHAIN_WS_CASE(144)
+                default:
+                    set_err("bad chain mfma K");
+                    return FSDR_ERR_INVALID;
+            }
+#undef CHAIN_WS_CASE
             HIP_TRY(hipGetLastError());
             return FSDR_OK;
         }
