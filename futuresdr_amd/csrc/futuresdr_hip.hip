@@ -4646,9 +4646,12 @@ extern "C" int fsdr_chain_run_dev(fsdr_chain* c, const void* d_in,
             return FSDR_OK;
         }
         const char* ws = getenv("FSDR_CHAIN_WS");
-        if (ws && atoi(ws) != 0 && L == 1024 &&
+        /* software-pipelined FFT-in-MFMA-shadow variant: measured ~4%
+         * faster than the ap kernel at fft_len 1024 (the serial FFT
+         * issues inside the MFMA pipe shadow) — default for the
+         * aligned 1024 case; FSDR_CHAIN_WS=0 disables. */
+        if ((!ws || atoi(ws) != 0) && L == 1024 &&
             ((uintptr_t)d_in & 15u) == 0) {
-            /* experiment: software-pipelined FFT-in-MFMA-shadow variant */
             unsigned SPm = (elemsP + 31u) & ~31u;
             size_t lds_ws = (8 * (size_t)SPm + 4 * ((size_t)KK + 16)) *
                                 sizeof(float) +
