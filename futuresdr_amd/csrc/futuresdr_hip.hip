@@ -1280,7 +1280,11 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
     };
     const unsigned abase = (unsigned)wave * 64 + 4u * r16;
     const unsigned asub = (unsigned)k4 * SUB;
-    auto mfma_vloc = [&](int v, v4f& cre, v4f& cim) {
+    /* g0/g1 select a sub-range of the vloc's b128 K-groups so the last
+     * vloc can be split across two FFT-stage windows (tail = run the
+     * non-multiple-of-4 K remainder too) */
+    auto mfma_vloc = [&](int v, v4f& cre, v4f& cim, int g0, int g1,
+                         bool tail) {
         const float* pre = planes + (unsigned)v * SPm + asub;
         const float* pim = planes + (4u + v) * SPm + asub;
         float bfrag[KKD / 4];
@@ -1290,6 +1294,7 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int t = 0; t < (KKD / 4) / 4; t++) {
+            if (t < g0 || t >= g1) continue;
             float4 ar = *(const float4*)&pre[abase + 4 * t];
             float4 ai = *(const float4*)&pim[abase + 4 * t];
             cre = __builtin_amdgcn_mfma_f32_16x16x4f32(ar.x, bfrag[4 * t],
@@ -1309,14 +1314,16 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
             cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
                 ai.w, bfrag[4 * t + 3], cim, 0, 0, 0);
         }
+        if (tail) {
 #pragma unroll
-        for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
-            float a_re = pre[abase + s];
-            float a_im = pim[abase + s];
-            cre = __builtin_amdgcn_mfma_f32_16x16x4f32(a_re, bfrag[s], cre,
-                                                       0, 0, 0);
-            cim = __builtin_amdgcn_mfma_f32_16x16x4f32(a_im, bfrag[s], cim,
-                                                       0, 0, 0);
+            for (int s = (KKD / 4) & ~3; s < KKD / 4; s++) {
+                float a_re = pre[abase + s];
+                float a_im = pim[abase + s];
+                cre = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_re, bfrag[s], cre, 0, 0, 0);
+                cim = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    a_im, bfrag[s], cim, 0, 0, 0);
+            }
         }
         __builtin_amdgcn_s_setprio(0);
     };
@@ -1366,21 +1373,24 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
             load_all(tile + gridDim.x);
         cre = (v4f){0.f, 0.f, 0.f, 0.f};
         cim = (v4f){0.f, 0.f, 0.f, 0.f};
-        mfma_vloc(0, cre, cim);
+        constexpr int G = (KKD / 4) / 4;
+        constexpr int GH = G / 2;
+        mfma_vloc(0, cre, cim, 0, G, true);
         if (have_prev) fft_stage(0);
         __syncthreads();
-        mfma_vloc(1, cre, cim);
+        mfma_vloc(1, cre, cim, 0, G, true);
         if (have_prev) fft_stage(1);
         __syncthreads();
-        mfma_vloc(2, cre, cim);
+        mfma_vloc(2, cre, cim, 0, G, true);
         if (have_prev) fft_stage(2);
         __syncthreads();
-        mfma_vloc(3, cre, cim);
+        mfma_vloc(3, cre, cim, 0, GH, false);
         if (have_prev) fft_stage(3);
         __syncthreads();
+        mfma_vloc(3, cre, cim, GH, G, true);
+        if (have_prev) fft_stage(4);
+        __syncthreads();
         if (have_prev) {
-            fft_stage(4);
-            __syncthreads();
             for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
                 long long o = prev_base + i;
                 if (o < n_out) {
