@@ -1226,12 +1226,8 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
 
     constexpr unsigned GROUPS = (MDFIR_TILE + KKD + 8) + 1;
     constexpr int NG = (GROUPS + MDFIR_BLOCK - 1) / MDFIR_BLOCK;
-    /* TWO staging register sets: tile t+2's loads are issued while t+1's
-     * sit ready, keeping two tile-loads in flight per block (6/CU at
-     * 3 blocks) — at this occupancy the VGPR budget is 170/wave, so the
-     * extra ~40 registers cost nothing. */
-    float4 stqa[2][NG], stqb[2][NG];
-    auto load_all = [&](long long tl, int setw) {
+    float4 stqa[NG], stqb[NG];
+    auto load_all = [&](long long tl) {
         const long long qb = tl * MDFIR_TILE;
 #pragma unroll
         for (int j = 0; j < NG; j++) {
@@ -1256,16 +1252,16 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
                     b = make_float4(e2.x, e2.y, e3.x, e3.y);
                 }
             }
-            stqa[setw][j] = a;
-            stqb[setw][j] = b;
+            stqa[j] = a;
+            stqb[j] = b;
         }
     };
-    auto write_all = [&](int setw) {
+    auto write_all = [&]() {
 #pragma unroll
         for (int j = 0; j < NG; j++) {
             unsigned q = (unsigned)(tid + j * MDFIR_BLOCK);
             if (q >= GROUPS) continue;
-            float4 a = stqa[setw][j], b = stqb[setw][j];
+            float4 a = stqa[j], b = stqb[j];
             if (q >= 1) {
                 unsigned d1 = ((q - 1) & 3u) * SUB + ((q - 1) >> 2);
                 planes[1u * SPm + d1] = a.x;
@@ -1360,13 +1356,10 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
     v4f cim = {0.f, 0.f, 0.f, 0.f};
     bool have_prev = false;
     long long prev_base = 0;
-    int par = 0;
-    load_all(blockIdx.x, 0);
-    if ((blockIdx.x + gridDim.x) * (long long)MDFIR_TILE < n_out)
-        load_all(blockIdx.x + gridDim.x, 1);
+    load_all(blockIdx.x);
     for (long long tile = blockIdx.x;
          tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
-        write_all(par);
+        write_all();
         if (have_prev) { /* deposit C(prev) into the FFT strip */
 #pragma unroll
             for (int q = 0; q < 4; q++) {
@@ -1376,8 +1369,8 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
             }
         }
         __syncthreads();
-        if ((tile + 2 * (long long)gridDim.x) * MDFIR_TILE < n_out)
-            load_all(tile + 2 * (long long)gridDim.x, par);
+        if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
+            load_all(tile + gridDim.x);
         cre = (v4f){0.f, 0.f, 0.f, 0.f};
         cim = (v4f){0.f, 0.f, 0.f, 0.f};
         constexpr int G = (KKD / 4) / 4;
@@ -1409,7 +1402,6 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
         }
         prev_base = tile * MDFIR_TILE;
         have_prev = true;
-        par ^= 1;
         __syncthreads();
     }
     /* epilogue: FFT + output of the final tile */
