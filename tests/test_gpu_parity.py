@@ -284,11 +284,12 @@ def test_fft_bluestein_inverse_shift_normalize(gpu, oracle_lib):
         assert rel < 2e-4, (kw, rel)
 
 
-def test_chain_allphase_vs_halves_kernels(gpu, oracle_lib):
-    """The two fused-kernel variants (all-phase aligned-group staging =
-    default for 16B-aligned inputs; 2-phase-halves = the unaligned ring
-    fallback) compute the same MFMA/K order — outputs must agree to
-    float equality on identical input."""
+def test_chain_kernel_variants_agree(gpu, oracle_lib):
+    """The three fused-kernel variants — ws (software-pipelined FFT,
+    the aligned-1024 default), ap (all-phase aligned-group staging) and
+    halves (the unaligned ring fallback) — compute the same MFMA/K and
+    butterfly order: outputs must agree to float equality on identical
+    input."""
     r = rng(4400)
     t1 = r.uniform(-1, 1, 127).astype(np.float32)
     t2 = r.uniform(-1, 1, 127).astype(np.float32)
@@ -299,21 +300,30 @@ def test_chain_allphase_vs_halves_kernels(gpu, oracle_lib):
     d_out = ctypes.c_void_p()
     assert lib.fsdr_dev_alloc(ctypes.byref(d_in), n_in * 8) == 0
     assert lib.fsdr_dev_alloc(ctypes.byref(d_out), n_in * 8) == 0
+    variants = {
+        "ws": {},
+        "ap": {"FSDR_CHAIN_WS": "0"},
+        "halves": {"FSDR_CHAIN_WS": "0", "FSDR_CHAIN_ALLPHASE": "0"},
+    }
     try:
         lib.fsdr_memcpy_h2d(d_in, ctypes.c_void_p(x.ctypes.data), n_in * 8)
         outs = {}
-        for env in ("1", "0"):
-            os.environ["FSDR_CHAIN_ALLPHASE"] = env
+        for name, env in variants.items():
+            for k in ("FSDR_CHAIN_WS", "FSDR_CHAIN_ALLPHASE"):
+                os.environ.pop(k, None)
+            os.environ.update(env)
             ch = gpu.Chain(t1, t2, 4, 1024)
             cons, prod = ch.run_dev(d_in.value, n_in, d_out.value, n_in)
             gpu.synchronize()
             h = np.zeros(prod, np.complex64)
             lib.fsdr_memcpy_d2h(ctypes.c_void_p(h.ctypes.data), d_out,
                                 prod * 8)
-            outs[env] = h
-        os.environ.pop("FSDR_CHAIN_ALLPHASE", None)
-        assert outs["1"].size == outs["0"].size
-        assert_close(outs["1"], outs["0"], 1e-6)
+            outs[name] = h
+        for k in ("FSDR_CHAIN_WS", "FSDR_CHAIN_ALLPHASE"):
+            os.environ.pop(k, None)
+        assert outs["ws"].size == outs["ap"].size == outs["halves"].size
+        assert_close(outs["ws"], outs["ap"], 1e-6)
+        assert_close(outs["ap"], outs["halves"], 1e-6)
     finally:
         lib.fsdr_dev_free(d_in)
         lib.fsdr_dev_free(d_out)
