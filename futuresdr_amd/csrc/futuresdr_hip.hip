@@ -1354,39 +1354,19 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
 
     v4f cre = {0.f, 0.f, 0.f, 0.f};
     v4f cim = {0.f, 0.f, 0.f, 0.f};
-    bool have_prev = false; /* C of the last-MFMA'd tile is pending */
-    bool have_out = false;  /* fbuf holds a finished spectrum */
-    long long prev_base = 0, out_base_p = 0;
-    auto output_fbuf = [&](long long ob) {
-        for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
-            long long o = ob + i;
-            if (o < n_out) {
-                float2 v = fbuf[fft_swz(rev4_10((unsigned)i))];
-                if (out) out[o] = v;
-                if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
-            }
-        }
-    };
-    auto deposit_c = [&]() {
-#pragma unroll
-        for (int q = 0; q < 4; q++) {
-            int row = k4 * 4 + q;
-            unsigned pos = wave * 256 + 16 * row + r16;
-            fbuf[fft_swz(pos)] = make_float2(cre[q], cim[q]);
-        }
-    };
+    bool have_prev = false;
+    long long prev_base = 0;
     load_all(blockIdx.x);
     for (long long tile = blockIdx.x;
          tile * (long long)MDFIR_TILE < n_out; tile += gridDim.x) {
-        /* the previous spectrum drains in the same window as the LDS
-         * staging writes (it used to be a serial window of its own) */
         write_all();
-        if (have_out) output_fbuf(out_base_p);
-        __syncthreads();
-        if (have_prev) { /* deposit C(prev) into the now-free strip */
-            deposit_c();
-            out_base_p = prev_base;
-            have_out = true;
+        if (have_prev) { /* deposit C(prev) into the FFT strip */
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                int row = k4 * 4 + q;
+                unsigned pos = wave * 256 + 16 * row + r16;
+                fbuf[fft_swz(pos)] = make_float2(cre[q], cim[q]);
+            }
         }
         __syncthreads();
         if ((tile + gridDim.x) * (long long)MDFIR_TILE < n_out)
@@ -1396,37 +1376,55 @@ __global__ __launch_bounds__(MDFIR_BLOCK) void k_decim4_fft_mfma_ws_tpl(
         constexpr int G = (KKD / 4) / 4;
         constexpr int GH = G / 2;
         mfma_vloc(0, cre, cim, 0, G, true);
-        if (have_out) fft_stage(0);
+        if (have_prev) fft_stage(0);
         __syncthreads();
         mfma_vloc(1, cre, cim, 0, G, true);
-        if (have_out) fft_stage(1);
+        if (have_prev) fft_stage(1);
         __syncthreads();
         mfma_vloc(2, cre, cim, 0, G, true);
-        if (have_out) fft_stage(2);
+        if (have_prev) fft_stage(2);
         __syncthreads();
         mfma_vloc(3, cre, cim, 0, GH, false);
-        if (have_out) fft_stage(3);
+        if (have_prev) fft_stage(3);
         __syncthreads();
         mfma_vloc(3, cre, cim, GH, G, true);
-        if (have_out) fft_stage(4);
+        if (have_prev) fft_stage(4);
+        __syncthreads();
+        if (have_prev) {
+            for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+                long long o = prev_base + i;
+                if (o < n_out) {
+                    float2 v = fbuf[fft_swz(rev4_10((unsigned)i))];
+                    if (out) out[o] = v;
+                    if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+                }
+            }
+        }
         prev_base = tile * MDFIR_TILE;
         have_prev = true;
         __syncthreads();
     }
-    /* epilogue: drain the pending spectrum, then FFT + output the
-     * final tile's C */
-    if (have_out) {
-        output_fbuf(out_base_p);
-        __syncthreads();
-    }
+    /* epilogue: FFT + output of the final tile */
     if (have_prev) {
-        deposit_c();
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            int row = k4 * 4 + q;
+            unsigned pos = wave * 256 + 16 * row + r16;
+            fbuf[fft_swz(pos)] = make_float2(cre[q], cim[q]);
+        }
         __syncthreads();
         for (int s = 0; s < 5; s++) {
             fft_stage(s);
             __syncthreads();
         }
-        output_fbuf(prev_base);
+        for (int i = tid; i < 1024; i += MDFIR_BLOCK) {
+            long long o = prev_base + i;
+            if (o < n_out) {
+                float2 v = fbuf[fft_swz(rev4_10((unsigned)i))];
+                if (out) out[o] = v;
+                if (mag_out) mag_out[o] = v.x * v.x + v.y * v.y;
+            }
+        }
     }
 }
 
